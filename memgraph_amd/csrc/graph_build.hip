@@ -1,0 +1,484 @@
+// Device-side graph construction for gfx950: deterministic edge-stream
+// generation, COO -> CSR (histogram + exclusive scan + scatter), and
+// degree-bin work lists for the fused sweep kernels.
+//
+// Replaces (MI355X-native, from scratch) the host-side layouts built by
+// pagerank_alg::PageRankGraph (reference algorithm/pagerank.cpp:166-182) and
+// GetGrappoloSuitableGraph (reference louvain.cpp:158-233; sym-CSR, every
+// edge stored twice) with int32 indices / fp32 weights in HBM.
+
+#include <cstring>
+
+#include <rocprim/rocprim.hpp>
+
+#include "../../include/mgx_graphgen.h"
+#include "mgx_internal.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+inline int64_t grid_for(int64_t work, int64_t cap = 4096) {
+  int64_t g = (work + kBlock - 1) / kBlock;
+  if (g < 1) g = 1;
+  if (g > cap) g = cap;
+  return g;
+}
+
+__global__ void k_rmat(int64_t n_edges, int scale, uint64_t mixed_seed,
+                       mgx_rmat_thresholds t, int32_t *src, int32_t *dst) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t s, d;
+    mgx_rmat_edge(mixed_seed, (uint64_t)i, scale, t, &s, &d);
+    src[i] = (int32_t)s;
+    dst[i] = (int32_t)d;
+  }
+}
+
+__global__ void k_uniform(int64_t n_edges, uint64_t mixed_seed, uint64_t n_vertices,
+                          int32_t *src, int32_t *dst) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t s, d;
+    mgx_uniform_edge(mixed_seed, (uint64_t)i, n_vertices, &s, &d);
+    src[i] = (int32_t)s;
+    dst[i] = (int32_t)d;
+  }
+}
+
+__global__ void k_weights(int64_t n_edges, uint64_t mixed_seed, float *w) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    w[i] = (float)mgx_edge_weight(mixed_seed, (uint64_t)i);
+  }
+}
+
+__global__ void k_zero_u32(int64_t n, uint32_t *p) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    p[i] = 0;
+}
+
+__global__ void k_hist(int64_t n_edges, const int32_t *idx, uint32_t *counts) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x)
+    atomicAdd(&counts[idx[i]], 1u);
+}
+
+__global__ void k_hist2(int64_t n_edges, const int32_t *a, const int32_t *b,
+                        uint32_t *counts) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    atomicAdd(&counts[a[i]], 1u);
+    atomicAdd(&counts[b[i]], 1u);
+  }
+}
+
+__global__ void k_hist_ranged(int64_t n_edges, const int32_t *idx, int32_t lo, int32_t hi,
+                              uint32_t *counts /* [hi-lo] */) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t v = idx[i];
+    if (v >= lo && v < hi) atomicAdd(&counts[v - lo], 1u);
+  }
+}
+
+__global__ void k_scatter_in(int64_t n_edges, const int32_t *src, const int32_t *dst,
+                             uint32_t *cursor, int32_t *col) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t pos = atomicAdd(&cursor[dst[i]], 1u);
+    col[pos] = src[i];
+  }
+}
+
+__global__ void k_scatter_in_ranged(int64_t n_edges, const int32_t *src, const int32_t *dst,
+                                    int32_t lo, int32_t hi, uint32_t *cursor, int32_t *col) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t v = dst[i];
+    if (v >= lo && v < hi) {
+      uint32_t pos = atomicAdd(&cursor[v - lo], 1u);
+      col[pos] = src[i];
+    }
+  }
+}
+
+__global__ void k_scatter_sym(int64_t n_edges, const int32_t *src, const int32_t *dst,
+                              const float *w, uint32_t *cursor, int32_t *col, float *out_w) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t s = src[i], d = dst[i];
+    const float wi = w ? w[i] : 1.0f;
+    uint32_t pos = atomicAdd(&cursor[s], 1u);
+    col[pos] = d;
+    if (out_w) out_w[pos] = wi;
+    pos = atomicAdd(&cursor[d], 1u);
+    col[pos] = s;
+    if (out_w) out_w[pos] = wi;
+  }
+}
+
+__global__ void k_inv_outdeg(int64_t n, const uint32_t *deg, float *inv) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    inv[i] = deg[i] ? 1.0f / (float)deg[i] : 0.0f;
+}
+
+__global__ void k_bin_keys(int64_t rows, const uint32_t *row_ptr, uint32_t *keys,
+                           uint32_t *vals) {
+  for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
+       r += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t deg = row_ptr[r + 1] - row_ptr[r];
+    uint32_t k = deg < 8 ? 0u : deg < 64 ? 1u : deg < 1024 ? 2u : 3u;
+    keys[r] = k;
+    vals[r] = (uint32_t)r;
+  }
+}
+
+__global__ void k_count_bins(int64_t rows, const uint32_t *keys, uint32_t *counts4) {
+  for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
+       r += (int64_t)gridDim.x * blockDim.x)
+    atomicAdd(&counts4[keys[r]], 1u);
+}
+
+__global__ void k_i64_to_i32(int64_t n, const int64_t *in, int32_t *out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int32_t)in[i];
+}
+
+__global__ void k_f64_to_f32(int64_t n, const double *in, float *out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (float)in[i];
+}
+
+// Exclusive scan of u32 counts[n] -> row_ptr[n+1] (row_ptr[n] = total).
+mgx_status scan_counts(mgx_context *ctx, const uint32_t *counts, int64_t n,
+                       uint32_t *row_ptr) {
+  size_t tmp_bytes = 0;
+  auto err = rocprim::exclusive_scan(nullptr, tmp_bytes, counts, row_ptr, 0u, n + 1,
+                                     rocprim::plus<uint32_t>(), ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  // Scan n+1 inputs (counts has n entries; read one past would be invalid) —
+  // instead scan n entries into row_ptr[0..n) and set the total separately.
+  err = rocprim::exclusive_scan(tmp, tmp_bytes, counts, row_ptr, 0u, n,
+                                rocprim::plus<uint32_t>(), ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  // row_ptr[n] = row_ptr[n-1] + counts[n-1]
+  uint32_t last_off = 0, last_cnt = 0;
+  if (n > 0) {
+    MGX_HIP_TRY(hipMemcpyAsync(&last_off, row_ptr + n - 1, 4, hipMemcpyDeviceToHost,
+                               ctx->stream));
+    MGX_HIP_TRY(hipMemcpyAsync(&last_cnt, counts + n - 1, 4, hipMemcpyDeviceToHost,
+                               ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  }
+  const uint32_t total = last_off + last_cnt;
+  MGX_HIP_TRY(hipMemcpyAsync(row_ptr + n, &total, 4, hipMemcpyHostToDevice, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  return MGX_OK;
+}
+
+}  // namespace
+
+mgx_status mgx_gen_rmat_device(mgx_context *ctx, int scale, int64_t n_edges, uint64_t seed,
+                               double a, double b, double c, int32_t *d_src, int32_t *d_dst) {
+  const uint64_t ms = mgx_seed_mix(seed);
+  const mgx_rmat_thresholds t = mgx_rmat_make_thresholds(a, b, c);
+  hipLaunchKernelGGL(k_rmat, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream, n_edges,
+                     scale, ms, t, d_src, d_dst);
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+mgx_status mgx_gen_uniform_device(mgx_context *ctx, int64_t n_vertices, int64_t n_edges,
+                                  uint64_t seed, int32_t *d_src, int32_t *d_dst) {
+  const uint64_t ms = mgx_seed_mix(seed);
+  hipLaunchKernelGGL(k_uniform, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                     n_edges, ms, (uint64_t)n_vertices, d_src, d_dst);
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+mgx_status mgx_gen_weights_device(mgx_context *ctx, int64_t n_edges, uint64_t seed,
+                                  float *d_w) {
+  const uint64_t ms = mgx_seed_mix(seed);
+  hipLaunchKernelGGL(k_weights, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                     n_edges, ms, d_w);
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+mgx_status mgx_build_bins(mgx_context *ctx, const uint32_t *row_ptr, int64_t rows,
+                          mgx_bins *bins) {
+  if (rows == 0) {
+    bins->rows = nullptr;
+    return MGX_OK;
+  }
+  MGX_HIP_TRY(hipMalloc(&bins->rows, rows * sizeof(int32_t)));
+  uint32_t *keys = nullptr, *keys_out = nullptr, *vals = nullptr;
+  MGX_HIP_TRY(hipMalloc(&keys, rows * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMalloc(&keys_out, rows * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMalloc(&vals, rows * sizeof(uint32_t)));
+  uint32_t *counts4 = nullptr;
+  MGX_HIP_TRY(hipMalloc(&counts4, 4 * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMemsetAsync(counts4, 0, 16, ctx->stream));
+
+  hipLaunchKernelGGL(k_bin_keys, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream, rows,
+                     row_ptr, keys, vals);
+  hipLaunchKernelGGL(k_count_bins, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream, rows,
+                     keys, counts4);
+
+  // Stable 2-bit radix sort: within a bin, rows stay in ascending id order.
+  size_t tmp_bytes = 0;
+  auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys, keys_out, vals,
+                                       (uint32_t *)bins->rows, rows, 0, 2, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::radix_sort_pairs(tmp, tmp_bytes, keys, keys_out, vals,
+                                  (uint32_t *)bins->rows, rows, 0, 2, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+
+  uint32_t h_counts[4] = {0, 0, 0, 0};
+  MGX_HIP_TRY(hipMemcpyAsync(h_counts, counts4, 16, hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  for (int b = 0; b < 4; ++b) bins->count[b] = h_counts[b];
+
+  // Launch geometry: rows-per-block {64,16,4,1}; grid-stride caps keep the
+  // dispatch bounded (guide G11) while >> 256 CUs stay fed.
+  const int64_t rows_per_block[4] = {64, 16, 4, 1};
+  const int64_t cap[4] = {2048, 2048, 2048, 8192};
+  for (int b = 0; b < 4; ++b) {
+    int64_t need = (bins->count[b] + rows_per_block[b] - 1) / rows_per_block[b];
+    bins->grid[b] = bins->count[b] ? (need < cap[b] ? need : cap[b]) : 0;
+  }
+
+  MGX_HIP_TRY(hipFree(keys));
+  MGX_HIP_TRY(hipFree(keys_out));
+  MGX_HIP_TRY(hipFree(vals));
+  MGX_HIP_TRY(hipFree(counts4));
+  return MGX_OK;
+}
+
+mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
+                                     const int32_t *d_dst, const float *d_w,
+                                     int64_t n_vertices, int64_t n_edges, uint32_t flags,
+                                     mgx_graph *g) {
+  const int64_t V = n_vertices, E = n_edges;
+  g->n_vertices = V;
+  g->n_edges = E;
+  g->flags = flags;
+  g->row_begin = 0;
+  g->row_end = V;
+  g->in_edges = E;
+
+  hipEvent_t ev0, ev1;
+  MGX_HIP_TRY(hipEventCreate(&ev0));
+  MGX_HIP_TRY(hipEventCreate(&ev1));
+  MGX_HIP_TRY(hipEventRecord(ev0, ctx->stream));
+
+  uint32_t *counts = nullptr;
+  MGX_HIP_TRY(hipMalloc(&counts, (V > 0 ? V : 1) * sizeof(uint32_t)));
+
+  // out-degree (+ inv) — always needed by PageRank/Katz result paths.
+  MGX_HIP_TRY(hipMalloc(&g->out_degree, (V > 0 ? V : 1) * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMemsetAsync(g->out_degree, 0, V * sizeof(uint32_t), ctx->stream));
+  hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
+                     g->out_degree);
+  MGX_HIP_TRY(hipMalloc(&g->inv_outdeg, (V > 0 ? V : 1) * sizeof(float)));
+  hipLaunchKernelGGL(k_inv_outdeg, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
+                     g->out_degree, g->inv_outdeg);
+
+  if (flags & MGX_BUILD_IN_CSR) {
+    MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
+    hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_dst,
+                       counts);
+    MGX_HIP_TRY(hipMalloc(&g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_TRY(scan_counts(ctx, counts, V, g->in_row_ptr));
+    MGX_HIP_TRY(hipMalloc(&g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
+    // cursor = copy of row_ptr[0..V)
+    MGX_HIP_TRY(hipMemcpyAsync(counts, g->in_row_ptr, V * sizeof(uint32_t),
+                               hipMemcpyDeviceToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_scatter_in, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
+                       d_src, d_dst, counts, g->in_col);
+    MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, V, &g->bins_in));
+  }
+
+  if (flags & MGX_BUILD_SYM_CSR) {
+    MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
+    hipLaunchKernelGGL(k_hist2, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
+                       d_dst, counts);
+    MGX_HIP_TRY(hipMalloc(&g->sym_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_TRY(scan_counts(ctx, counts, V, g->sym_row_ptr));
+    MGX_HIP_TRY(hipMalloc(&g->sym_col, (E > 0 ? 2 * E : 1) * sizeof(int32_t)));
+    if (flags & MGX_BUILD_WEIGHTED) {
+      MGX_HIP_TRY(hipMalloc(&g->sym_w, (E > 0 ? 2 * E : 1) * sizeof(float)));
+    }
+    MGX_HIP_TRY(hipMemcpyAsync(counts, g->sym_row_ptr, V * sizeof(uint32_t),
+                               hipMemcpyDeviceToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_scatter_sym, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
+                       d_src, d_dst, d_w, counts, g->sym_col, g->sym_w);
+    MGX_TRY(mgx_build_bins(ctx, g->sym_row_ptr, V, &g->bins_sym));
+  }
+
+  MGX_HIP_TRY(hipFree(counts));
+  MGX_HIP_TRY(hipEventRecord(ev1, ctx->stream));
+  MGX_HIP_TRY(hipEventSynchronize(ev1));
+  float ms = 0.f;
+  MGX_HIP_TRY(hipEventElapsedTime(&ms, ev0, ev1));
+  g->build_ms = ms;
+  MGX_HIP_TRY(hipEventDestroy(ev0));
+  MGX_HIP_TRY(hipEventDestroy(ev1));
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+mgx_status mgx_build_sharded_in_csr(mgx_context *ctx, const int32_t *d_src,
+                                    const int32_t *d_dst, int64_t n_vertices,
+                                    int64_t n_edges, int64_t row_begin, int64_t row_end,
+                                    mgx_graph *g) {
+  const int64_t V = n_vertices, E = n_edges;
+  const int64_t rows = row_end - row_begin;
+  g->n_vertices = V;
+  g->n_edges = E;
+  g->flags = MGX_BUILD_IN_CSR;
+  g->row_begin = row_begin;
+  g->row_end = row_end;
+
+  hipEvent_t ev0, ev1;
+  MGX_HIP_TRY(hipEventCreate(&ev0));
+  MGX_HIP_TRY(hipEventCreate(&ev1));
+  MGX_HIP_TRY(hipEventRecord(ev0, ctx->stream));
+
+  // Global out-degree (contrib denominators need every source).
+  MGX_HIP_TRY(hipMalloc(&g->out_degree, V * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMemsetAsync(g->out_degree, 0, V * sizeof(uint32_t), ctx->stream));
+  hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
+                     g->out_degree);
+  MGX_HIP_TRY(hipMalloc(&g->inv_outdeg, V * sizeof(float)));
+  hipLaunchKernelGGL(k_inv_outdeg, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
+                     g->out_degree, g->inv_outdeg);
+
+  uint32_t *counts = nullptr;
+  MGX_HIP_TRY(hipMalloc(&counts, (rows > 0 ? rows : 1) * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMemsetAsync(counts, 0, rows * sizeof(uint32_t), ctx->stream));
+  hipLaunchKernelGGL(k_hist_ranged, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
+                     d_dst, (int32_t)row_begin, (int32_t)row_end, counts);
+  MGX_HIP_TRY(hipMalloc(&g->in_row_ptr, (rows + 1) * sizeof(uint32_t)));
+  MGX_TRY(scan_counts(ctx, counts, rows, g->in_row_ptr));
+  uint32_t local_edges = 0;
+  MGX_HIP_TRY(hipMemcpyAsync(&local_edges, g->in_row_ptr + rows, 4, hipMemcpyDeviceToHost,
+                             ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  g->in_edges = local_edges;
+  MGX_HIP_TRY(hipMalloc(&g->in_col, (local_edges > 0 ? local_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(hipMemcpyAsync(counts, g->in_row_ptr, rows * sizeof(uint32_t),
+                             hipMemcpyDeviceToDevice, ctx->stream));
+  hipLaunchKernelGGL(k_scatter_in_ranged, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream,
+                     E, d_src, d_dst, (int32_t)row_begin, (int32_t)row_end, counts,
+                     g->in_col);
+  MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, rows, &g->bins_in));
+  MGX_HIP_TRY(hipFree(counts));
+
+  MGX_HIP_TRY(hipEventRecord(ev1, ctx->stream));
+  MGX_HIP_TRY(hipEventSynchronize(ev1));
+  float ms = 0.f;
+  MGX_HIP_TRY(hipEventElapsedTime(&ms, ev0, ev1));
+  g->build_ms = ms;
+  MGX_HIP_TRY(hipEventDestroy(ev0));
+  MGX_HIP_TRY(hipEventDestroy(ev1));
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+// Host COO (int64/double) upload helpers, used by mgx_graph_from_coo.
+mgx_status mgx_upload_coo(mgx_context *ctx, const int64_t *src, const int64_t *dst,
+                          const double *weights, int64_t n_edges, int32_t **d_src,
+                          int32_t **d_dst, float **d_w) {
+  MGX_HIP_TRY(hipMalloc(d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(hipMalloc(d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  *d_w = nullptr;
+  if (weights) MGX_HIP_TRY(hipMalloc(d_w, (n_edges > 0 ? n_edges : 1) * sizeof(float)));
+
+  // Chunked staging: int64 -> int32 converted on device.
+  const int64_t chunk = 16 << 20;
+  int64_t *stage64 = nullptr;
+  double *stagef = nullptr;
+  const int64_t this_chunk = n_edges < chunk ? n_edges : chunk;
+  if (this_chunk > 0) {
+    MGX_HIP_TRY(hipMalloc(&stage64, this_chunk * sizeof(int64_t)));
+    if (weights) MGX_HIP_TRY(hipMalloc(&stagef, this_chunk * sizeof(double)));
+  }
+  for (int64_t off = 0; off < n_edges; off += chunk) {
+    const int64_t n = (n_edges - off) < chunk ? (n_edges - off) : chunk;
+    MGX_HIP_TRY(hipMemcpyAsync(stage64, src + off, n * sizeof(int64_t),
+                               hipMemcpyHostToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_i64_to_i32, dim3(grid_for(n)), dim3(kBlock), 0, ctx->stream, n,
+                       stage64, *d_src + off);
+    MGX_HIP_TRY(hipMemcpyAsync(stage64, dst + off, n * sizeof(int64_t),
+                               hipMemcpyHostToDevice, ctx->stream));
+    hipLaunchKernelGGL(k_i64_to_i32, dim3(grid_for(n)), dim3(kBlock), 0, ctx->stream, n,
+                       stage64, *d_dst + off);
+    if (weights) {
+      MGX_HIP_TRY(hipMemcpyAsync(stagef, weights + off, n * sizeof(double),
+                                 hipMemcpyHostToDevice, ctx->stream));
+      hipLaunchKernelGGL(k_f64_to_f32, dim3(grid_for(n)), dim3(kBlock), 0, ctx->stream, n,
+                         stagef, *d_w + off);
+    }
+    // The next chunk reuses the staging buffer: wait for queued copies.
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  }
+  if (stage64) MGX_HIP_TRY(hipFree(stage64));
+  if (stagef) MGX_HIP_TRY(hipFree(stagef));
+  return MGX_OK;
+}
+
+namespace {
+__global__ void k_i32_to_i64(int64_t n, const int32_t *in, int64_t *out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int64_t)in[i];
+}
+}  // namespace
+
+// Test support: device-generate an edge list and download it (int64), so
+// tests can check bit-identity with the numpy/C generators.
+extern "C" mgx_status mgx_gen_edges_to_host(mgx_context *ctx, int rmat, int scale,
+                                            int64_t n_vertices, int64_t n_edges,
+                                            uint64_t seed, double a, double b, double c,
+                                            int64_t *out_src, int64_t *out_dst) {
+  MGX_HIP_TRY(hipSetDevice(ctx->device));
+  int32_t *d_src = nullptr, *d_dst = nullptr;
+  int64_t *d_wide = nullptr;
+  MGX_HIP_TRY(hipMalloc(&d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(hipMalloc(&d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(hipMalloc(&d_wide, (n_edges > 0 ? n_edges : 1) * sizeof(int64_t)));
+  mgx_status s;
+  if (rmat) {
+    s = mgx_gen_rmat_device(ctx, scale, n_edges, seed, a, b, c, d_src, d_dst);
+  } else {
+    s = mgx_gen_uniform_device(ctx, n_vertices, n_edges, seed, d_src, d_dst);
+  }
+  if (s == MGX_OK) {
+    hipLaunchKernelGGL(k_i32_to_i64, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                       n_edges, d_src, d_wide);
+    (void)hipMemcpyAsync(out_src, d_wide, n_edges * sizeof(int64_t), hipMemcpyDeviceToHost,
+                         ctx->stream);
+    (void)hipStreamSynchronize(ctx->stream);
+    hipLaunchKernelGGL(k_i32_to_i64, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                       n_edges, d_dst, d_wide);
+    (void)hipMemcpyAsync(out_dst, d_wide, n_edges * sizeof(int64_t), hipMemcpyDeviceToHost,
+                         ctx->stream);
+    (void)hipStreamSynchronize(ctx->stream);
+  }
+  (void)hipFree(d_src);
+  (void)hipFree(d_dst);
+  (void)hipFree(d_wide);
+  return s;
+}

@@ -1,0 +1,404 @@
+// PageRank on gfx950 — replaces pagerank_alg::ParallelIterativePageRank
+// (reference algorithm/pagerank.cpp:194-242) with a fused pull-SpMV sweep:
+//
+//   new_rank[v] = (1-d)/N + d * sum_{u->v} contrib[u],  contrib[u] = rank[u]/outdeg(u)
+//
+// One kernel launch per iteration covers all degree bins (sub-wave work
+// assignment per mgx_bins): f32 ranks/contrib in HBM, f64 accumulation in
+// registers, and the same kernel writes new_rank AND new_contrib and folds
+// the per-block Linf(delta) into a device scalar — so per-iteration traffic
+// is the algorithmic 8 B/edge + ~20 B/vertex (DESIGN.md roofline).
+//
+// Parity bar (tests/test_gpu_pagerank.py): |r_gpu - r_cpu|inf <= 1e-6 vs the
+// fp64 oracle after equal iterations, post sum-normalize.
+
+#include <chrono>
+
+#include "mgx_internal.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+struct PrArgs {
+  const uint32_t *row_ptr;
+  const int32_t *col;
+  const int32_t *bin_rows;
+  int64_t n[4];     // rows per bin
+  int64_t off[4];   // offsets into bin_rows
+  int64_t goff[4];  // grid offsets per section
+  int64_t grid[4];
+  const float *contrib_old;
+  const float *rank_old;
+  float *rank_new;
+  float *contrib_new;
+  const float *inv_outdeg;  // global vertex id
+  int64_t row_base;         // sharded: global row = row_base + local row
+  float base_term;          // (1-d)/N
+  float damping;
+  uint32_t *delta_max;      // f32-as-ordered-uint
+};
+
+template <int LANES>
+__device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) {
+  constexpr int RPB = kBlock / LANES;
+  const int64_t nrows = A.n[sec];
+  const int32_t *rows_list = A.bin_rows + A.off[sec];
+  const int sub = threadIdx.x % LANES;
+  float maxd = 0.0f;
+  __shared__ double red[4];
+  for (int64_t base = block_in_sec * RPB; base < nrows; base += A.grid[sec] * RPB) {
+    const int64_t ri = base + threadIdx.x / LANES;
+    double acc = 0.0;
+    int32_t row = -1;
+    if (ri < nrows) {
+      row = rows_list[ri];
+      const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
+      for (uint32_t j = s + sub; j < e; j += LANES) acc += (double)A.contrib_old[A.col[j]];
+    }
+    if constexpr (LANES <= 64) {
+      for (int o = LANES / 2; o; o >>= 1) acc += __shfl_down(acc, o, LANES);
+    } else {
+      for (int o = 32; o; o >>= 1) acc += __shfl_down(acc, o, 64);
+      if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+      __syncthreads();
+      if (threadIdx.x == 0) acc = red[0] + red[1] + red[2] + red[3];
+    }
+    if (sub == 0 && row >= 0 && (LANES <= 64 || threadIdx.x == 0)) {
+      const float newr = (float)((double)A.base_term + (double)A.damping * acc);
+      const int64_t gv = A.row_base + row;
+      const float oldr = A.rank_old[gv];
+      A.rank_new[gv] = newr;
+      A.contrib_new[gv] = newr * A.inv_outdeg[gv];
+      const float d = fabsf(newr - oldr);
+      if (d > maxd) maxd = d;
+    }
+    if constexpr (LANES > 64) __syncthreads();
+  }
+  return maxd;
+}
+
+__global__ void __launch_bounds__(kBlock) k_pr_sweep(PrArgs A) {
+  const int64_t b = blockIdx.x;
+  int sec = 3;
+  if (b < A.goff[1]) sec = 0;
+  else if (b < A.goff[2]) sec = 1;
+  else if (b < A.goff[3]) sec = 2;
+  const int64_t bis = b - A.goff[sec];
+
+  float maxd;
+  switch (sec) {
+    case 0: maxd = pr_rows<4>(A, 0, bis); break;
+    case 1: maxd = pr_rows<16>(A, 1, bis); break;
+    case 2: maxd = pr_rows<64>(A, 2, bis); break;
+    default: maxd = pr_rows<256>(A, 3, bis); break;
+  }
+
+  if (A.delta_max) {
+    // block max -> one atomic per block (guide G12).
+    __shared__ float wmax[kBlock / 64];
+    for (int o = 32; o; o >>= 1) maxd = fmaxf(maxd, __shfl_down(maxd, o, 64));
+    if ((threadIdx.x & 63) == 0) wmax[threadIdx.x >> 6] = maxd;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float m = wmax[0];
+      for (int i = 1; i < kBlock / 64; ++i) m = fmaxf(m, wmax[i]);
+      atomicMax(A.delta_max, __float_as_uint(m));
+    }
+  }
+}
+
+__global__ void k_pr_init(int64_t n, float r0, const float *inv, float *rank,
+                          float *contrib) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    rank[i] = r0;
+    contrib[i] = r0 * inv[i];
+  }
+}
+
+__global__ void k_sum_f32(int64_t n, const float *x, double *out) {
+  double acc = 0.0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    acc += (double)x[i];
+  __shared__ double red[kBlock / 64];
+  for (int o = 32; o; o >>= 1) acc += __shfl_down(acc, o, 64);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double s = 0.0;
+    for (int i = 0; i < kBlock / 64; ++i) s += red[i];
+    atomicAdd(out, s);
+  }
+}
+
+__global__ void k_pr_widen(int64_t n, const float *rank, const double *sum, double *out) {
+  const double inv = 1.0 / *sum;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (double)rank[i] * inv;
+}
+
+inline int64_t grid_for(int64_t work, int64_t cap = 4096) {
+  int64_t g = (work + kBlock - 1) / kBlock;
+  if (g < 1) g = 1;
+  return g > cap ? cap : g;
+}
+
+mgx_status queue_one_iteration(mgx_pagerank_run *run, bool track_delta) {
+  mgx_context *ctx = run->ctx;
+  mgx_graph *g = run->g;
+  const int64_t V = g->n_vertices;
+
+  PrArgs A;
+  A.row_ptr = g->in_row_ptr;
+  A.col = g->in_col;
+  A.bin_rows = g->bins_in.rows;
+  int64_t off = 0, goff = 0;
+  for (int b = 0; b < 4; ++b) {
+    A.n[b] = g->bins_in.count[b];
+    A.off[b] = off;
+    off += A.n[b];
+    A.goff[b] = goff;
+    A.grid[b] = g->bins_in.grid[b];
+    goff += A.grid[b];
+  }
+  A.contrib_old = run->contrib[run->cur];
+  A.rank_old = run->rank[run->cur];
+  A.rank_new = run->rank[1 - run->cur];
+  A.contrib_new = run->contrib[1 - run->cur];
+  A.inv_outdeg = g->inv_outdeg;
+  A.row_base = g->row_begin;
+  A.base_term = (float)((1.0 - run->damping) / (double)V);
+  A.damping = (float)run->damping;
+  A.delta_max = track_delta ? run->d_delta : nullptr;
+
+  if (track_delta) MGX_HIP_TRY(hipMemsetAsync(run->d_delta, 0, 4, ctx->stream));
+
+  const int64_t total_grid = goff;
+  if (total_grid > 0 && V > 0) {
+    // Event-bracket the sweep (the dominant kernel): feeds roofline.achieved.
+    if (run->ev_used >= (int64_t)run->ev_start.size()) {
+      hipEvent_t e0, e1;
+      MGX_HIP_TRY(hipEventCreate(&e0));
+      MGX_HIP_TRY(hipEventCreate(&e1));
+      run->ev_start.push_back(e0);
+      run->ev_stop.push_back(e1);
+    }
+    MGX_HIP_TRY(hipEventRecord(run->ev_start[run->ev_used], ctx->stream));
+    hipLaunchKernelGGL(k_pr_sweep, dim3((uint32_t)total_grid), dim3(kBlock), 0, ctx->stream,
+                       A);
+    MGX_HIP_TRY(hipEventRecord(run->ev_stop[run->ev_used], ctx->stream));
+    ++run->ev_used;
+    MGX_HIP_TRY(hipGetLastError());
+  }
+
+  // Distributed: every rank owns rows [row_begin,row_end); exchange the new
+  // rank+contrib slices so the next gather sees all sources
+  // (ncclAllGather over xGMI).
+  if (run->dist) {
+    const int world = mgx_comm_world(ctx);
+    const int64_t shard = run->row_end - run->row_begin;  // equal on all ranks (padded)
+    MGX_TRY(mgx_comm_allgather_f32(ctx, A.contrib_new + run->row_begin, A.contrib_new,
+                                   (size_t)shard));
+    MGX_TRY(mgx_comm_allgather_f32(ctx, A.rank_new + run->row_begin, A.rank_new,
+                                   (size_t)shard));
+    (void)world;
+  }
+
+  run->cur = 1 - run->cur;
+  ++run->iterations;
+  // Bounded event ring: fold timings once in a while.
+  if (run->ev_used >= 1024) MGX_TRY(run->flush_timing());
+  return MGX_OK;
+}
+
+}  // namespace
+
+mgx_status mgx_pagerank_run::flush_timing() {
+  if (ev_used == 0) return MGX_OK;
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  for (int64_t i = 0; i < ev_used; ++i) {
+    float ms = 0.f;
+    MGX_HIP_TRY(hipEventElapsedTime(&ms, ev_start[i], ev_stop[i]));
+    sweep_ms_acc += ms;
+  }
+  launches_acc += ev_used;
+  ev_used = 0;
+  return MGX_OK;
+}
+
+mgx_status mgx_pagerank_queue_iterations(mgx_pagerank_run *run, int64_t n, bool track_delta) {
+  for (int64_t i = 0; i < n; ++i) MGX_TRY(queue_one_iteration(run, track_delta));
+  return MGX_OK;
+}
+
+mgx_status mgx_pagerank_read_delta(mgx_pagerank_run *run, float *out) {
+  uint32_t bits = 0;
+  MGX_HIP_TRY(hipMemcpyAsync(&bits, run->d_delta, 4, hipMemcpyDeviceToHost,
+                             run->ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(run->ctx->stream));
+  union {
+    uint32_t u;
+    float f;
+  } cv;
+  cv.u = bits;
+  *out = cv.f;
+  return MGX_OK;
+}
+
+mgx_status mgx_pagerank_normalize_download(mgx_pagerank_run *run, double *out_rank) {
+  mgx_context *ctx = run->ctx;
+  const int64_t V = run->g->n_vertices;
+  if (V == 0) return MGX_OK;
+  // NormalizeRank (reference pagerank.cpp:157-162): divide by the sum.
+  MGX_HIP_TRY(hipMemsetAsync(run->d_scratch, 0, sizeof(double), ctx->stream));
+  hipLaunchKernelGGL(k_sum_f32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
+                     V, run->rank[run->cur], run->d_scratch);
+  hipLaunchKernelGGL(k_pr_widen, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
+                     V, run->rank[run->cur], run->d_scratch, run->d_scratch + 1);
+  if (out_rank) {
+    MGX_HIP_TRY(hipMemcpyAsync(out_rank, run->d_scratch + 1, V * sizeof(double),
+                               hipMemcpyDeviceToHost, ctx->stream));
+  }
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  return MGX_OK;
+}
+
+namespace {
+
+mgx_status pagerank_start_common(mgx_context *ctx, mgx_graph *g, double damping,
+                                 bool dist, int64_t row_begin, int64_t row_end,
+                                 mgx_pagerank_run **out) {
+  if (!ctx || !g || !(g->flags & MGX_BUILD_IN_CSR)) {
+    mgx_set_error("pagerank needs a graph built with MGX_BUILD_IN_CSR");
+    return MGX_ERR_INVALID_ARGUMENT;
+  }
+  auto *run = new mgx_pagerank_run();
+  run->ctx = ctx;
+  run->g = g;
+  run->damping = damping;
+  run->dist = dist;
+  run->row_begin = row_begin;
+  run->row_end = row_end;
+  const int64_t V = g->n_vertices;
+  // Distributed rank/contrib arrays are padded to world*shard.
+  const int64_t alloc = dist ? (row_end - row_begin) * (int64_t)mgx_comm_world(ctx) : V;
+  const int64_t n = (alloc > V ? alloc : V);
+  for (int i = 0; i < 2; ++i) {
+    MGX_HIP_TRY(hipMalloc(&run->rank[i], (n > 0 ? n : 1) * sizeof(float)));
+    MGX_HIP_TRY(hipMalloc(&run->contrib[i], (n > 0 ? n : 1) * sizeof(float)));
+  }
+  MGX_HIP_TRY(hipMalloc(&run->d_delta, sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMalloc(&run->d_scratch, (V + 1) * sizeof(double)));
+  if (V > 0) {
+    const float r0 = (float)(1.0 / (double)V);
+    hipLaunchKernelGGL(k_pr_init, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
+                       V, r0, g->inv_outdeg, run->rank[0], run->contrib[0]);
+    MGX_HIP_TRY(hipGetLastError());
+  }
+  *out = run;
+  return MGX_OK;
+}
+
+void pagerank_run_free(mgx_pagerank_run *run) {
+  for (int i = 0; i < 2; ++i) {
+    if (run->rank[i]) (void)hipFree(run->rank[i]);
+    if (run->contrib[i]) (void)hipFree(run->contrib[i]);
+  }
+  if (run->d_delta) (void)hipFree(run->d_delta);
+  if (run->d_scratch) (void)hipFree(run->d_scratch);
+  for (auto e : run->ev_start) (void)hipEventDestroy(e);
+  for (auto e : run->ev_stop) (void)hipEventDestroy(e);
+  delete run;
+}
+
+}  // namespace
+
+extern "C" mgx_status mgx_pagerank_start(mgx_context *ctx, mgx_graph *g, double damping,
+                                         mgx_pagerank_run **out) {
+  return pagerank_start_common(ctx, g, damping, false, 0, g ? g->n_vertices : 0, out);
+}
+
+extern "C" mgx_status mgx_pagerank_start_dist(mgx_context *ctx, mgx_graph *g, double damping,
+                                              int64_t row_begin, int64_t row_end,
+                                              mgx_pagerank_run **out) {
+  if (mgx_comm_world(ctx) == 0) {
+    mgx_set_error("mgx_comm_init must be called before mgx_pagerank_start_dist");
+    return MGX_ERR_INVALID_ARGUMENT;
+  }
+  return pagerank_start_common(ctx, g, damping, true, row_begin, row_end, out);
+}
+
+extern "C" mgx_status mgx_pagerank_iterate(mgx_pagerank_run *run, int64_t n) {
+  return mgx_pagerank_queue_iterations(run, n, /*track_delta=*/false);
+}
+
+extern "C" mgx_status mgx_pagerank_timing(mgx_pagerank_run *run, double *sweep_ms,
+                                          int64_t *launches) {
+  MGX_TRY(run->flush_timing());
+  if (sweep_ms) *sweep_ms = run->sweep_ms_acc;
+  if (launches) *launches = run->launches_acc;
+  return MGX_OK;
+}
+
+extern "C" mgx_status mgx_pagerank_finish(mgx_pagerank_run *run, double *out_rank) {
+  mgx_status s = mgx_pagerank_normalize_download(run, out_rank);
+  if (s == MGX_OK) s = run->flush_timing();
+  pagerank_run_free(run);
+  return s;
+}
+
+extern "C" mgx_status mgx_pagerank(mgx_context *ctx, mgx_graph *g, int64_t max_iterations,
+                                   double damping, double stop_epsilon, double *out_rank,
+                                   mgx_pagerank_stats *stats) {
+  mgx_pagerank_run *run = nullptr;
+  MGX_TRY(mgx_pagerank_start(ctx, g, damping, &run));
+  hipEvent_t it0, it1;
+  MGX_HIP_TRY(hipEventCreate(&it0));
+  MGX_HIP_TRY(hipEventCreate(&it1));
+  MGX_HIP_TRY(hipEventRecord(it0, ctx->stream));
+
+  mgx_status s = MGX_OK;
+  if (stop_epsilon > 0.0) {
+    // Reference stopping rule (pagerank.cpp:139-151): after each iteration,
+    // stop when Linf(new-old) <= eps or the cap is reached.
+    for (int64_t i = 0; i < max_iterations && s == MGX_OK; ++i) {
+      s = mgx_pagerank_queue_iterations(run, 1, /*track_delta=*/true);
+      if (s != MGX_OK) break;
+      float delta = 0.f;
+      s = mgx_pagerank_read_delta(run, &delta);
+      if (delta <= (float)stop_epsilon) break;
+    }
+  } else {
+    s = mgx_pagerank_queue_iterations(run, max_iterations, false);
+  }
+
+  MGX_HIP_TRY(hipEventRecord(it1, ctx->stream));
+
+  double download_ms = 0.0;
+  if (s == MGX_OK) {
+    MGX_HIP_TRY(hipEventSynchronize(it1));
+    const auto t0 = std::chrono::steady_clock::now();
+    s = mgx_pagerank_normalize_download(run, out_rank);
+    const auto t1 = std::chrono::steady_clock::now();
+    download_ms = std::chrono::duration<double, std::milli>(t1 - t0).count();
+  }
+
+  if (stats && s == MGX_OK) {
+    float iter_ms = 0.f;
+    MGX_HIP_TRY(hipEventElapsedTime(&iter_ms, it0, it1));
+    (void)run->flush_timing();
+    stats->iterations = run->iterations;
+    stats->iter_ms = iter_ms;
+    stats->sweep_ms = run->sweep_ms_acc;
+    stats->sweep_launches = run->launches_acc;
+    stats->csr_build_ms = g->build_ms;
+    stats->download_ms = download_ms;
+  }
+  MGX_HIP_TRY(hipEventDestroy(it0));
+  MGX_HIP_TRY(hipEventDestroy(it1));
+  pagerank_run_free(run);
+  return s;
+}

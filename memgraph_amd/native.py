@@ -1,0 +1,200 @@
+"""ctypes bindings for libmgx_analytics.so (the gfx950 compute library).
+
+Product-path plumbing for bench.py and the GPU tests. Fails loudly when the
+library is missing or no HIP device is present — there is no CPU fallback.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+LIB_PATH = os.path.join(_DIR, "lib", "libmgx_analytics.so")
+
+BUILD_IN_CSR = 1
+BUILD_SYM_CSR = 2
+BUILD_WEIGHTED = 4
+
+UNIQUE_ID_BYTES = 128
+
+_I64 = ctypes.POINTER(ctypes.c_int64)
+_F64 = ctypes.POINTER(ctypes.c_double)
+
+
+class PagerankStats(ctypes.Structure):
+    _fields_ = [
+        ("iterations", ctypes.c_int64),
+        ("iter_ms", ctypes.c_double),
+        ("sweep_ms", ctypes.c_double),
+        ("sweep_launches", ctypes.c_int64),
+        ("csr_build_ms", ctypes.c_double),
+        ("download_ms", ctypes.c_double),
+    ]
+
+
+class MgxError(RuntimeError):
+    pass
+
+
+class Native:
+    def __init__(self, path=LIB_PATH):
+        if not os.path.exists(path):
+            raise MgxError(f"{path} not built — run `make` / __graft_entry__.build()")
+        self.lib = ctypes.CDLL(path)
+        self.lib.mgx_status_string.restype = ctypes.c_char_p
+        self.lib.mgx_last_error.restype = ctypes.c_char_p
+        self.lib.mgx_graph_build_ms.restype = ctypes.c_double
+        self.lib.mgx_graph_num_vertices.restype = ctypes.c_int64
+        self.lib.mgx_graph_num_edges.restype = ctypes.c_int64
+
+    def _check(self, status, what):
+        if status != 0:
+            msg = self.lib.mgx_status_string(status).decode()
+            detail = self.lib.mgx_last_error().decode()
+            raise MgxError(f"{what}: {msg} ({detail})")
+
+    def device_count(self):
+        return self.lib.mgx_device_count()
+
+    def init(self, device=0):
+        ctx = ctypes.c_void_p()
+        self._check(self.lib.mgx_init(device, ctypes.byref(ctx)), "mgx_init")
+        return ctx
+
+    def destroy(self, ctx):
+        self.lib.mgx_destroy(ctx)
+
+    def sync(self, ctx):
+        self._check(self.lib.mgx_sync(ctx), "mgx_sync")
+
+    # --- graphs ---
+    def graph_from_coo(self, ctx, src, dst, n_vertices, weights=None, flags=BUILD_IN_CSR):
+        src = np.ascontiguousarray(src, dtype=np.int64)
+        dst = np.ascontiguousarray(dst, dtype=np.int64)
+        pw = None
+        if weights is not None:
+            weights = np.ascontiguousarray(weights, dtype=np.float64)
+            pw = weights.ctypes.data_as(_F64)
+        g = ctypes.c_void_p()
+        self._check(
+            self.lib.mgx_graph_from_coo(ctx, src.ctypes.data_as(_I64),
+                                        dst.ctypes.data_as(_I64), pw,
+                                        ctypes.c_int64(n_vertices),
+                                        ctypes.c_int64(len(src)), ctypes.c_uint32(flags),
+                                        ctypes.byref(g)), "mgx_graph_from_coo")
+        return g
+
+    def graph_rmat(self, ctx, scale, n_edges, seed=1, a=0.57, b=0.19, c=0.19,
+                   flags=BUILD_IN_CSR, weight_seed=7):
+        g = ctypes.c_void_p()
+        self._check(
+            self.lib.mgx_graph_rmat(ctx, ctypes.c_int(scale), ctypes.c_int64(n_edges),
+                                    ctypes.c_uint64(seed), ctypes.c_double(a),
+                                    ctypes.c_double(b), ctypes.c_double(c),
+                                    ctypes.c_uint32(flags), ctypes.c_uint64(weight_seed),
+                                    ctypes.byref(g)), "mgx_graph_rmat")
+        return g
+
+    def graph_uniform(self, ctx, n_vertices, n_edges, seed=42, flags=BUILD_IN_CSR,
+                      weight_seed=7):
+        g = ctypes.c_void_p()
+        self._check(
+            self.lib.mgx_graph_uniform(ctx, ctypes.c_int64(n_vertices),
+                                       ctypes.c_int64(n_edges), ctypes.c_uint64(seed),
+                                       ctypes.c_uint32(flags), ctypes.c_uint64(weight_seed),
+                                       ctypes.byref(g)), "mgx_graph_uniform")
+        return g
+
+    def graph_rmat_sharded(self, ctx, scale, n_edges, row_begin, row_end, seed=1, a=0.57,
+                           b=0.19, c=0.19):
+        g = ctypes.c_void_p()
+        self._check(
+            self.lib.mgx_graph_rmat_sharded(ctx, ctypes.c_int(scale),
+                                            ctypes.c_int64(n_edges), ctypes.c_uint64(seed),
+                                            ctypes.c_double(a), ctypes.c_double(b),
+                                            ctypes.c_double(c), ctypes.c_int64(row_begin),
+                                            ctypes.c_int64(row_end), ctypes.byref(g)),
+            "mgx_graph_rmat_sharded")
+        return g
+
+    def graph_destroy(self, ctx, g):
+        self.lib.mgx_graph_destroy(ctx, g)
+
+    def graph_build_ms(self, g):
+        return self.lib.mgx_graph_build_ms(g)
+
+    # --- algorithms ---
+    def pagerank(self, ctx, g, n_vertices, max_iterations=100, damping=0.85, eps=1e-5):
+        out = np.zeros(n_vertices, dtype=np.float64)
+        stats = PagerankStats()
+        self._check(
+            self.lib.mgx_pagerank(ctx, g, ctypes.c_int64(max_iterations),
+                                  ctypes.c_double(damping), ctypes.c_double(eps),
+                                  out.ctypes.data_as(_F64), ctypes.byref(stats)),
+            "mgx_pagerank")
+        return out, stats
+
+    def pagerank_start(self, ctx, g, damping=0.85):
+        run = ctypes.c_void_p()
+        self._check(self.lib.mgx_pagerank_start(ctx, g, ctypes.c_double(damping),
+                                                ctypes.byref(run)), "mgx_pagerank_start")
+        return run
+
+    def pagerank_start_dist(self, ctx, g, row_begin, row_end, damping=0.85):
+        run = ctypes.c_void_p()
+        self._check(
+            self.lib.mgx_pagerank_start_dist(ctx, g, ctypes.c_double(damping),
+                                             ctypes.c_int64(row_begin),
+                                             ctypes.c_int64(row_end), ctypes.byref(run)),
+            "mgx_pagerank_start_dist")
+        return run
+
+    def pagerank_iterate(self, run, n):
+        self._check(self.lib.mgx_pagerank_iterate(run, ctypes.c_int64(n)),
+                    "mgx_pagerank_iterate")
+
+    def pagerank_timing(self, run):
+        ms = ctypes.c_double()
+        n = ctypes.c_int64()
+        self._check(self.lib.mgx_pagerank_timing(run, ctypes.byref(ms), ctypes.byref(n)),
+                    "mgx_pagerank_timing")
+        return ms.value, n.value
+
+    def pagerank_finish(self, run, n_vertices=0, want_rank=True):
+        out = np.zeros(n_vertices, dtype=np.float64) if want_rank else None
+        p = out.ctypes.data_as(_F64) if want_rank and n_vertices else None
+        self._check(self.lib.mgx_pagerank_finish(run, p), "mgx_pagerank_finish")
+        return out
+
+    def wcc(self, ctx, g, n_vertices):
+        out = np.zeros(n_vertices, dtype=np.int64)
+        n = ctypes.c_int64()
+        self._check(self.lib.mgx_wcc(ctx, g, out.ctypes.data_as(_I64), ctypes.byref(n)),
+                    "mgx_wcc")
+        return out, n.value
+
+    def katz(self, ctx, g, n_vertices, alpha=0.2, epsilon=1e-2):
+        out = np.zeros(n_vertices, dtype=np.float64)
+        iters = ctypes.c_int64()
+        self._check(
+            self.lib.mgx_katz(ctx, g, ctypes.c_double(alpha), ctypes.c_double(epsilon),
+                              out.ctypes.data_as(_F64), ctypes.byref(iters)), "mgx_katz")
+        return out, iters.value
+
+    def louvain(self, ctx, g, n_vertices, threshold=1e-6):
+        out = np.zeros(n_vertices, dtype=np.int64)
+        n = ctypes.c_int64()
+        self._check(
+            self.lib.mgx_louvain(ctx, g, ctypes.c_double(threshold),
+                                 out.ctypes.data_as(_I64), ctypes.byref(n)), "mgx_louvain")
+        return out, n.value
+
+    # --- comm ---
+    def comm_unique_id(self):
+        buf = (ctypes.c_char * UNIQUE_ID_BYTES)()
+        self._check(self.lib.mgx_comm_unique_id(buf), "mgx_comm_unique_id")
+        return bytes(buf)
+
+    def comm_init(self, ctx, rank, world, id_bytes):
+        buf = (ctypes.c_char * UNIQUE_ID_BYTES).from_buffer_copy(id_bytes)
+        self._check(self.lib.mgx_comm_init(ctx, rank, world, buf), "mgx_comm_init")
