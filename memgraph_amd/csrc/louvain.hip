@@ -1,18 +1,989 @@
 // Louvain community detection on gfx950 — replaces the grappolo basic path
-// the reference community_detection module runs (runMultiPhaseBasic.cpp
-// :53-146 + parallelLouvainMethod.cpp:65-290). Implementation lands in this
-// round after the SpMV-family kernels are validated; until then the entry
-// point reports NOT_SUPPORTED (never a CPU fallback).
+// the reference community_detection module runs with coloring=false:
+//   - Jacobi sweep: parallelLouvianMethod (grappolo
+//     BasicCommunitiesDetection/parallelLouvainMethod.cpp:65-290) — per
+//     vertex, aggregate neighbour-community weights, pick the max-dQ
+//     community with grappolo's exact rule (Utility/
+//     utilityClusteringFunctions.cpp max():275-310: strict argmax, nonzero
+//     ties to the smaller community id, singleton-swap protection), update
+//     community degree/size deltas, modularity from the PRE-update state;
+//     result is the assignment BEFORE the last sweep (pastCommAss), and the
+//     empty-adjacency -1 target propagates through the triple rotation —
+//     all replicated (see oracle/src/louvain_oracle.cpp for the pinned
+//     sequential restatement).
+//   - renumberClustersContiguously (Utility/buildNextPhase.cpp:49-78):
+//     first-seen order == ascending order of each community's minimum
+//     vertex index — computed with atomicMin + a stable sort.
+//   - coarsening buildNextLevelGraphOpt (Utility/buildNextPhase.cpp:82-):
+//     (C[i] >= C[tail]) pair aggregation by sort + reduce_by_key, self-loop
+//     entry always present, mirrored lower pairs.
+//   - phase loop runMultiPhaseBasic (BasicCommunitiesDetection/
+//     runMultiPhaseBasic.cpp:53-146).
+//
+// Device layout: per-level sym CSR (u32 row_ptr, i32 col, f64 weights —
+// level 0 widened from the graph's f32), community-keyed open-addressing
+// hash tables (LDS per wave for rows with deg < 256, a global pool with one
+// power-of-two region per hub row otherwise).
+//
+// Known divergence (documented, tests pin partitions on goldens +
+// modularity at tolerance elsewhere): e_xx and community-degree updates use
+// fp64 atomics, so summation order differs from the sequential oracle —
+// sub-1e-12 modularity noise that can only matter within 1e-12 of the
+// stopping threshold.
+
+#include <cstring>
+
+#include <rocprim/rocprim.hpp>
 
 #include "mgx_internal.h"
 
+namespace {
+
+constexpr int kBlock = 256;
+constexpr uint32_t kSmallRowDeg = 256;  // <: wave+LDS table; >=: block+pool
+constexpr int kLdsCap = 512;            // per-wave table entries (>= 2*255)
+
+inline int64_t grid_for(int64_t work, int64_t cap = 4096) {
+  int64_t g = (work + kBlock - 1) / kBlock;
+  if (g < 1) g = 1;
+  return g > cap ? cap : g;
+}
+
+struct Level {
+  int64_t nv = 0;
+  uint32_t *row_ptr = nullptr;  // [nv+1]
+  int32_t *col = nullptr;       // [ne2]
+  double *w = nullptr;          // [ne2]
+  int64_t ne2 = 0;              // entries (each input edge twice)
+};
+
+// ---------- generic small kernels ----------------------------------------
+
+__global__ void k_f32_to_f64(int64_t n, const float *in, double *out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (double)in[i];
+}
+
+__global__ void k_row_wsum(int64_t nv, const uint32_t *row_ptr, const double *w,
+                           double *vdeg, double *cinfo_deg, int32_t *cinfo_size) {
+  // sumVertexDegree (utilityClusteringFunctions.cpp:68-85).
+  for (int64_t v = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; v < nv;
+       v += (int64_t)gridDim.x * blockDim.x) {
+    double s = 0.0;
+    for (uint32_t j = row_ptr[v]; j < row_ptr[v + 1]; ++j) s += w[j];
+    vdeg[v] = s;
+    cinfo_deg[v] = s;
+    cinfo_size[v] = 1;
+  }
+}
+
+__global__ void k_sum_f64(int64_t n, const double *x, double *out) {
+  double acc = 0.0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    acc += x[i];
+  __shared__ double red[kBlock / 64];
+  for (int o = 32; o; o >>= 1) acc += __shfl_down(acc, o, 64);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double s = 0.0;
+    for (int i = 0; i < kBlock / 64; ++i) s += red[i];
+    atomicAdd(out, s);
+  }
+}
+
+__global__ void k_sum_sq_f64(int64_t n, const double *x, double *out) {
+  double acc = 0.0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    acc += x[i] * x[i];
+  __shared__ double red[kBlock / 64];
+  for (int o = 32; o; o >>= 1) acc += __shfl_down(acc, o, 64);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double s = 0.0;
+    for (int i = 0; i < kBlock / 64; ++i) s += red[i];
+    atomicAdd(out, s);
+  }
+}
+
+__global__ void k_iota_i32(int64_t n, int32_t *p) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    p[i] = (int32_t)i;
+}
+
+__global__ void k_fill_i32(int64_t n, int32_t v, int32_t *p) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    p[i] = v;
+}
+
+__global__ void k_apply_updates(int64_t nv, double *cinfo_deg, int32_t *cinfo_size,
+                                double *cupd_deg, int32_t *cupd_size) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    cinfo_deg[i] += cupd_deg[i];
+    cinfo_size[i] += cupd_size[i];
+    cupd_deg[i] = 0.0;
+    cupd_size[i] = 0;
+  }
+}
+
+// ---------- the sweep -----------------------------------------------------
+
+struct SweepArgs {
+  const uint32_t *row_ptr;
+  const int32_t *col;
+  const double *w;
+  const double *vdeg;
+  const double *cinfo_deg;
+  const int32_t *cinfo_size;
+  const int32_t *curr;
+  int32_t *target;
+  double *cupd_deg;
+  int32_t *cupd_size;
+  double *e_xx;  // += Counter[0] per vertex (clusterWeightInternal sum)
+  double constant;
+  // work lists
+  const int32_t *small_rows;
+  int64_t n_small;
+  const int32_t *big_rows;
+  int64_t n_big;
+  // global hash pool for big rows
+  int32_t *pool_keys;
+  double *pool_vals;
+  const uint64_t *pool_off;  // [n_big+1]
+};
+
+__device__ inline void wave_lds_fence() {
+  // Wave-level LDS completion + compiler ordering: the per-wave hash table
+  // region is touched only by this wave's 64 lanes, so draining lgkmcnt
+  // after the write phase makes every lane's LDS writes visible to every
+  // lane's subsequent reads (no cross-wave traffic => no s_barrier needed).
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
+__device__ inline double dq_gain(double eiy, double eix, double vdeg_i, double ay,
+                                 double ax, double constant) {
+  return 2.0 * (eiy - eix) - 2.0 * vdeg_i * (ay - ax) * constant;
+}
+
+// Deterministic argmax with grappolo's rule: strictly positive best gain,
+// ties to the smallest community id; otherwise stay.
+struct Best {
+  double gain;
+  int32_t cid;
+};
+
+__device__ inline Best best_merge(Best a, Best b) {
+  if (b.gain > a.gain) return b;
+  if (b.gain == a.gain && b.cid < a.cid) return b;
+  return a;
+}
+
+// One wave processes one small row using its LDS table region.
+__global__ void __launch_bounds__(kBlock) k_sweep_small(SweepArgs A) {
+  __shared__ int32_t keys[kBlock / 64][kLdsCap];
+  __shared__ double vals[kBlock / 64][kLdsCap];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t waves_per_grid = (int64_t)gridDim.x * (kBlock / 64);
+  for (int64_t wi = (int64_t)blockIdx.x * (kBlock / 64) + wave; wi < A.n_small;
+       wi += waves_per_grid) {
+    const int32_t row = A.small_rows[wi];
+    const uint32_t adj1 = A.row_ptr[row], adj2 = A.row_ptr[row + 1];
+    const int32_t sc = A.curr[row];
+    // clear table
+    for (int s = lane; s < kLdsCap; s += 64) {
+      keys[wave][s] = -1;
+      vals[wave][s] = 0.0;
+    }
+    wave_lds_fence();
+    // seed own community (buildLocalMapCounter seeds Counter[0]=0)
+    if (lane == 0) {
+      uint32_t h = ((uint32_t)sc * 2654435761u) & (kLdsCap - 1);
+      keys[wave][h] = sc;  // first insert, empty table: no probe needed
+    }
+    wave_lds_fence();
+    double self_loop = 0.0;
+    for (uint32_t j = adj1 + lane; j < adj2; j += 64) {
+      const int32_t nb = A.col[j];
+      const double wj = A.w[j];
+      if (nb == row) self_loop += wj;
+      const int32_t cid = A.curr[nb];
+      uint32_t h = ((uint32_t)cid * 2654435761u) & (kLdsCap - 1);
+      while (true) {
+        int32_t prev = atomicCAS(&keys[wave][h], -1, cid);
+        if (prev == -1 || prev == cid) break;
+        h = (h + 1) & (kLdsCap - 1);
+      }
+      atomicAdd(&vals[wave][h], wj);
+    }
+    for (int o = 32; o; o >>= 1) self_loop += __shfl_down(self_loop, o, 64);
+    self_loop = __shfl(self_loop, 0, 64);
+    wave_lds_fence();
+    // lookup own bucket
+    double own = 0.0;
+    {
+      uint32_t h = ((uint32_t)sc * 2654435761u) & (kLdsCap - 1);
+      while (keys[wave][h] != sc) h = (h + 1) & (kLdsCap - 1);
+      own = vals[wave][h];
+    }
+    const double eix = own - self_loop;
+    const double ax = A.cinfo_deg[sc] - A.vdeg[row];
+    Best best{0.0, sc};
+    for (int s = lane; s < kLdsCap; s += 64) {
+      const int32_t cid = keys[wave][s];
+      if (cid >= 0 && cid != sc) {
+        const double gain =
+            dq_gain(vals[wave][s], eix, A.vdeg[row], A.cinfo_deg[cid], ax, A.constant);
+        if (gain > best.gain || (gain == best.gain && gain != 0.0 && cid < best.cid))
+          best = Best{gain, cid};
+      }
+    }
+    for (int o = 32; o; o >>= 1) {
+      Best other{__shfl_down(best.gain, o, 64), __shfl_down(best.cid, o, 64)};
+      if (other.gain > best.gain ||
+          (other.gain == best.gain && other.gain != 0.0 && other.cid < best.cid))
+        best = other;
+    }
+    if (lane == 0) {
+      int32_t max_index = (best.gain > 0.0) ? best.cid : sc;
+      // swap protection (max(), utilityClusteringFunctions.cpp:305-307)
+      if (A.cinfo_size[max_index] == 1 && A.cinfo_size[sc] == 1 && max_index > sc)
+        max_index = sc;
+      A.target[row] = max_index;
+      if (max_index != sc) {
+        atomicAdd(&A.cupd_deg[max_index], A.vdeg[row]);
+        atomicAdd(&A.cupd_size[max_index], 1);
+        atomicAdd(&A.cupd_deg[sc], -A.vdeg[row]);
+        atomicAdd(&A.cupd_size[sc], -1);
+      }
+      atomicAdd(A.e_xx, own);
+    }
+    wave_lds_fence();
+  }
+}
+
+// One 256-thread block per big row, table region in the global pool.
+__global__ void __launch_bounds__(kBlock) k_sweep_big(SweepArgs A) {
+  __shared__ double s_red[kBlock / 64];
+  __shared__ int32_t s_cid[kBlock / 64];
+  for (int64_t bi = blockIdx.x; bi < A.n_big; bi += gridDim.x) {
+    const int32_t row = A.big_rows[bi];
+    const uint32_t adj1 = A.row_ptr[row], adj2 = A.row_ptr[row + 1];
+    const int32_t sc = A.curr[row];
+    const uint64_t t0 = A.pool_off[bi], t1 = A.pool_off[bi + 1];
+    const uint32_t cap = (uint32_t)(t1 - t0);  // power of two
+    int32_t *keys = A.pool_keys + t0;
+    double *vals = A.pool_vals + t0;
+    // pool was cleared before the sweep
+    if (threadIdx.x == 0) {
+      uint32_t h = ((uint32_t)sc * 2654435761u) & (cap - 1);
+      keys[h] = sc;
+    }
+    __syncthreads();
+    double self_loop = 0.0;
+    for (uint32_t j = adj1 + threadIdx.x; j < adj2; j += kBlock) {
+      const int32_t nb = A.col[j];
+      const double wj = A.w[j];
+      if (nb == row) self_loop += wj;
+      const int32_t cid = A.curr[nb];
+      uint32_t h = ((uint32_t)cid * 2654435761u) & (cap - 1);
+      while (true) {
+        int32_t prev = atomicCAS(&keys[h], -1, cid);
+        if (prev == -1 || prev == cid) break;
+        h = (h + 1) & (cap - 1);
+      }
+      atomicAdd(&vals[h], wj);
+    }
+    // block sum of self_loop
+    for (int o = 32; o; o >>= 1) self_loop += __shfl_down(self_loop, o, 64);
+    if ((threadIdx.x & 63) == 0) s_red[threadIdx.x >> 6] = self_loop;
+    __syncthreads();
+    double sl = s_red[0] + s_red[1] + s_red[2] + s_red[3];
+    __syncthreads();
+    double own = 0.0;
+    {
+      uint32_t h = ((uint32_t)sc * 2654435761u) & (cap - 1);
+      while (keys[h] != sc) h = (h + 1) & (cap - 1);
+      own = vals[h];
+    }
+    const double eix = own - sl;
+    const double ax = A.cinfo_deg[sc] - A.vdeg[row];
+    Best best{0.0, sc};
+    for (uint32_t s = threadIdx.x; s < cap; s += kBlock) {
+      const int32_t cid = keys[s];
+      if (cid >= 0 && cid != sc) {
+        const double gain =
+            dq_gain(vals[s], eix, A.vdeg[row], A.cinfo_deg[cid], ax, A.constant);
+        if (gain > best.gain || (gain == best.gain && gain != 0.0 && cid < best.cid))
+          best = Best{gain, cid};
+      }
+    }
+    for (int o = 32; o; o >>= 1) {
+      Best other{__shfl_down(best.gain, o, 64), __shfl_down(best.cid, o, 64)};
+      if (other.gain > best.gain ||
+          (other.gain == best.gain && other.gain != 0.0 && other.cid < best.cid))
+        best = other;
+    }
+    if ((threadIdx.x & 63) == 0) {
+      s_red[threadIdx.x >> 6] = best.gain;
+      s_cid[threadIdx.x >> 6] = best.cid;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      for (int i = 1; i < kBlock / 64; ++i) {
+        Best other{s_red[i], s_cid[i]};
+        if (other.gain > best.gain ||
+            (other.gain == best.gain && other.gain != 0.0 && other.cid < best.cid))
+          best = other;
+      }
+      int32_t max_index = (best.gain > 0.0) ? best.cid : sc;
+      if (A.cinfo_size[max_index] == 1 && A.cinfo_size[sc] == 1 && max_index > sc)
+        max_index = sc;
+      A.target[row] = max_index;
+      if (max_index != sc) {
+        atomicAdd(&A.cupd_deg[max_index], A.vdeg[row]);
+        atomicAdd(&A.cupd_size[max_index], 1);
+        atomicAdd(&A.cupd_deg[sc], -A.vdeg[row]);
+        atomicAdd(&A.cupd_size[sc], -1);
+      }
+      atomicAdd(A.e_xx, own);
+    }
+    __syncthreads();
+  }
+}
+
+// Rows with no edges: target = -1 (parallelLouvainMethod.cpp:209-211).
+__global__ void k_sweep_empty(int64_t nv, const uint32_t *row_ptr, int32_t *target) {
+  for (int64_t v = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; v < nv;
+       v += (int64_t)gridDim.x * blockDim.x) {
+    if (row_ptr[v] == row_ptr[v + 1]) target[v] = -1;
+  }
+}
+
+// ---------- renumber ------------------------------------------------------
+
+__global__ void k_rep_min(int64_t nv, const int32_t *C, int32_t *rep) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (C[i] >= 0) atomicMin(&rep[C[i]], (int32_t)i);
+  }
+}
+
+__global__ void k_newid_from_sorted(int64_t n_active, const int32_t *sorted_c,
+                                    int32_t *newid) {
+  for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_active;
+       k += (int64_t)gridDim.x * blockDim.x)
+    newid[sorted_c[k]] = (int32_t)k;
+}
+
+__global__ void k_remap(int64_t nv, int32_t *C, const int32_t *newid) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (C[i] >= 0) C[i] = newid[C[i]];
+  }
+}
+
+// ---------- coarsening ----------------------------------------------------
+
+__global__ void k_pair_keys(int64_t ne2, const int32_t *col, const uint32_t *row_ptr,
+                            int64_t nv, const int32_t *C, const double *w,
+                            const int32_t *row_of_entry, uint64_t *keys, double *vals,
+                            uint64_t sentinel) {
+  for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < ne2;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t i = row_of_entry[j];
+    const int32_t ci = C[i];
+    const int32_t ct = C[col[j]];
+    if (ci >= ct) {
+      keys[j] = ((uint64_t)(uint32_t)ci << 32) | (uint32_t)ct;
+      vals[j] = w[j];
+    } else {
+      keys[j] = sentinel;  // sorts last, dropped
+      vals[j] = 0.0;
+    }
+  }
+}
+
+__global__ void k_row_of_entry(int64_t nv, const uint32_t *row_ptr, int32_t *row_of_entry) {
+  for (int64_t v = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; v < nv;
+       v += (int64_t)gridDim.x * blockDim.x) {
+    for (uint32_t j = row_ptr[v]; j < row_ptr[v + 1]; ++j) row_of_entry[j] = (int32_t)v;
+  }
+}
+
+// count CSR rows for unique pairs (ci>=ct): row ci +1; if ct<ci also row ct +1.
+__global__ void k_pair_counts(int64_t n_pairs, const uint64_t *keys, uint32_t *counts,
+                              uint32_t *self_present) {
+  for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_pairs;
+       k += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t ci = (int32_t)(keys[k] >> 32);
+    const int32_t ct = (int32_t)(uint32_t)keys[k];
+    atomicAdd(&counts[ci], 1u);
+    if (ct < ci) atomicAdd(&counts[ct], 1u);
+    else self_present[ci] = 1u;  // ct == ci
+  }
+}
+
+__global__ void k_add_missing_self(int64_t ncl, const uint32_t *self_present,
+                                   uint32_t *counts) {
+  for (int64_t c = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; c < ncl;
+       c += (int64_t)gridDim.x * blockDim.x) {
+    if (!self_present[c]) atomicAdd(&counts[c], 1u);
+  }
+}
+
+__global__ void k_pair_scatter(int64_t n_pairs, const uint64_t *keys, const double *vals,
+                               uint32_t *cursor, int32_t *col, double *w) {
+  for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_pairs;
+       k += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t ci = (int32_t)(keys[k] >> 32);
+    const int32_t ct = (int32_t)(uint32_t)keys[k];
+    uint32_t p = atomicAdd(&cursor[ci], 1u);
+    col[p] = ct;
+    w[p] = vals[k];
+    if (ct < ci) {
+      p = atomicAdd(&cursor[ct], 1u);
+      col[p] = ci;
+      w[p] = vals[k];
+    }
+  }
+}
+
+__global__ void k_self_scatter(int64_t ncl, const uint32_t *self_present, uint32_t *cursor,
+                               int32_t *col, double *w) {
+  for (int64_t c = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; c < ncl;
+       c += (int64_t)gridDim.x * blockDim.x) {
+    if (!self_present[c]) {
+      uint32_t p = atomicAdd(&cursor[c], 1u);
+      col[p] = (int32_t)c;
+      w[p] = 0.0;  // zero-weight self loop (buildNextPhase.cpp cluPtrIn init)
+    }
+  }
+}
+
+// big-row pool capacities (power of two >= 2*deg)
+__global__ void k_big_caps(int64_t n_big, const int32_t *big_rows, const uint32_t *row_ptr,
+                           uint64_t *caps) {
+  for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_big;
+       k += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t row = big_rows[k];
+    uint64_t need = 2ull * (row_ptr[row + 1] - row_ptr[row]) + 2;
+    uint64_t cap = 1;
+    while (cap < need) cap <<= 1;
+    caps[k] = cap;
+  }
+}
+
+__global__ void k_classify_rows(int64_t nv, const uint32_t *row_ptr, uint32_t *n_small,
+                                uint32_t *n_big) {
+  for (int64_t v = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; v < nv;
+       v += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t deg = row_ptr[v + 1] - row_ptr[v];
+    if (deg == 0) continue;
+    if (deg < kSmallRowDeg) atomicAdd(n_small, 1u);
+    else atomicAdd(n_big, 1u);
+  }
+}
+
+__global__ void k_fill_rows(int64_t nv, const uint32_t *row_ptr, int32_t *small_rows,
+                            int32_t *big_rows, uint32_t *c_small, uint32_t *c_big) {
+  // Order within the lists is nondeterministic (atomics) but irrelevant:
+  // the sweep is a Jacobi update over a fixed snapshot.
+  for (int64_t v = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; v < nv;
+       v += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t deg = row_ptr[v + 1] - row_ptr[v];
+    if (deg == 0) continue;
+    if (deg < kSmallRowDeg) small_rows[atomicAdd(c_small, 1u)] = (int32_t)v;
+    else big_rows[atomicAdd(c_big, 1u)] = (int32_t)v;
+  }
+}
+
+// ---------- host orchestration -------------------------------------------
+
+struct DevBuf {
+  void *p = nullptr;
+  ~DevBuf() {
+    if (p) (void)hipFree(p);
+  }
+  hipError_t alloc(size_t bytes) { return hipMalloc(&p, bytes ? bytes : 1); }
+  template <typename T>
+  T *as() {
+    return (T *)p;
+  }
+};
+
+mgx_status read_scalar_f64(mgx_context *ctx, const double *d, double *out) {
+  MGX_HIP_TRY(hipMemcpyAsync(out, d, sizeof(double), hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  return MGX_OK;
+}
+
+// Runs parallelLouvianMethod on one level. C (device i32[nv]) receives
+// pastCommAss. Returns modularity via *out_mod.
+mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double thresh,
+                         int32_t *C, double *out_mod, int64_t *iters_out) {
+  const int64_t nv = L.nv;
+  DevBuf vdeg, cinfo_deg, cinfo_size, cupd_deg, cupd_size, past, curr, target, scalars;
+  MGX_HIP_TRY(vdeg.alloc(nv * 8));
+  MGX_HIP_TRY(cinfo_deg.alloc(nv * 8));
+  MGX_HIP_TRY(cinfo_size.alloc(nv * 4));
+  MGX_HIP_TRY(cupd_deg.alloc(nv * 8));
+  MGX_HIP_TRY(cupd_size.alloc(nv * 4));
+  MGX_HIP_TRY(past.alloc(nv * 4));
+  MGX_HIP_TRY(curr.alloc(nv * 4));
+  MGX_HIP_TRY(target.alloc(nv * 4));
+  MGX_HIP_TRY(scalars.alloc(3 * 8));  // [e_xx, a2_x, total_w]
+
+  hipLaunchKernelGGL(k_row_wsum, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, L.row_ptr, L.w, vdeg.as<double>(), cinfo_deg.as<double>(),
+                     cinfo_size.as<int32_t>());
+  MGX_HIP_TRY(hipMemsetAsync(scalars.as<double>() + 2, 0, 8, ctx->stream));
+  hipLaunchKernelGGL(k_sum_f64, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, vdeg.as<double>(), scalars.as<double>() + 2);
+  double total_w = 0.0;
+  MGX_TRY(read_scalar_f64(ctx, scalars.as<double>() + 2, &total_w));
+  const double constant = 1.0 / total_w;  // calConstantForSecondTerm
+
+  hipLaunchKernelGGL(k_iota_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, past.as<int32_t>());
+  hipLaunchKernelGGL(k_iota_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, curr.as<int32_t>());
+  MGX_HIP_TRY(hipMemsetAsync(cupd_deg.p, 0, nv * 8, ctx->stream));
+  MGX_HIP_TRY(hipMemsetAsync(cupd_size.p, 0, nv * 4, ctx->stream));
+
+  // Row classification + big-row pool.
+  DevBuf counters, small_rows, big_rows, pool_off, pool_keys, pool_vals;
+  MGX_HIP_TRY(counters.alloc(4 * 4));
+  MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
+  hipLaunchKernelGGL(k_classify_rows, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                     ctx->stream, nv, L.row_ptr, counters.as<uint32_t>(),
+                     counters.as<uint32_t>() + 1);
+  uint32_t h_counts[2] = {0, 0};
+  MGX_HIP_TRY(hipMemcpyAsync(h_counts, counters.p, 8, hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  const int64_t n_small = h_counts[0], n_big = h_counts[1];
+  MGX_HIP_TRY(small_rows.alloc(n_small * 4));
+  MGX_HIP_TRY(big_rows.alloc(n_big * 4));
+  MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
+  hipLaunchKernelGGL(k_fill_rows, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                     ctx->stream, nv, L.row_ptr, small_rows.as<int32_t>(),
+                     big_rows.as<int32_t>(), counters.as<uint32_t>(),
+                     counters.as<uint32_t>() + 1);
+  uint64_t pool_total = 0;
+  if (n_big > 0) {
+    DevBuf caps;
+    MGX_HIP_TRY(caps.alloc(n_big * 8));
+    hipLaunchKernelGGL(k_big_caps, dim3((uint32_t)grid_for(n_big)), dim3(kBlock), 0,
+                       ctx->stream, n_big, big_rows.as<int32_t>(), L.row_ptr,
+                       caps.as<uint64_t>());
+    MGX_HIP_TRY(pool_off.alloc((n_big + 1) * 8));
+    size_t tmp_bytes = 0;
+    auto err = rocprim::exclusive_scan(nullptr, tmp_bytes, caps.as<uint64_t>(),
+                                       pool_off.as<uint64_t>(), (uint64_t)0, n_big,
+                                       rocprim::plus<uint64_t>(), ctx->stream);
+    if (err != hipSuccess) return MGX_ERR_HIP;
+    void *tmp = nullptr;
+    MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+    err = rocprim::exclusive_scan(tmp, tmp_bytes, caps.as<uint64_t>(),
+                                  pool_off.as<uint64_t>(), (uint64_t)0, n_big,
+                                  rocprim::plus<uint64_t>(), ctx->stream);
+    if (err != hipSuccess) return MGX_ERR_HIP;
+    uint64_t last_off = 0, last_cap = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&last_off, pool_off.as<uint64_t>() + n_big - 1, 8,
+                               hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipMemcpyAsync(&last_cap, caps.as<uint64_t>() + n_big - 1, 8,
+                               hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    pool_total = last_off + last_cap;
+    MGX_HIP_TRY(hipMemcpyAsync(pool_off.as<uint64_t>() + n_big, &pool_total, 8,
+                               hipMemcpyHostToDevice, ctx->stream));
+    MGX_HIP_TRY(pool_keys.alloc(pool_total * 4));
+    MGX_HIP_TRY(pool_vals.alloc(pool_total * 8));
+  }
+
+  SweepArgs A;
+  A.row_ptr = L.row_ptr;
+  A.col = L.col;
+  A.w = L.w;
+  A.vdeg = vdeg.as<double>();
+  A.cinfo_deg = cinfo_deg.as<double>();
+  A.cinfo_size = cinfo_size.as<int32_t>();
+  A.cupd_deg = cupd_deg.as<double>();
+  A.cupd_size = cupd_size.as<int32_t>();
+  A.e_xx = scalars.as<double>();
+  A.constant = constant;
+  A.small_rows = small_rows.as<int32_t>();
+  A.n_small = n_small;
+  A.big_rows = big_rows.as<int32_t>();
+  A.n_big = n_big;
+  A.pool_keys = pool_keys.as<int32_t>();
+  A.pool_vals = pool_vals.as<double>();
+  A.pool_off = pool_off.as<uint64_t>();
+
+  double prev_mod = -1.0, curr_mod = -1.0;
+  int64_t iters = 0;
+  int32_t *p_past = past.as<int32_t>(), *p_curr = curr.as<int32_t>(),
+          *p_target = target.as<int32_t>();
+  while (true) {
+    ++iters;
+    MGX_HIP_TRY(hipMemsetAsync(scalars.p, 0, 16, ctx->stream));  // e_xx, a2_x
+    if (n_big > 0) {
+      hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for((int64_t)pool_total)),
+                         dim3(kBlock), 0, ctx->stream, (int64_t)pool_total, -1,
+                         pool_keys.as<int32_t>());
+      MGX_HIP_TRY(hipMemsetAsync(pool_vals.p, 0, pool_total * 8, ctx->stream));
+    }
+    A.curr = p_curr;
+    A.target = p_target;
+    hipLaunchKernelGGL(k_sweep_empty, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                       ctx->stream, nv, L.row_ptr, p_target);
+    if (n_small > 0) {
+      const int64_t waves_needed = n_small;
+      const int64_t blocks = grid_for(waves_needed * 64, 4096);
+      hipLaunchKernelGGL(k_sweep_small, dim3((uint32_t)blocks), dim3(kBlock), 0,
+                         ctx->stream, A);
+    }
+    if (n_big > 0) {
+      const int64_t blocks = n_big < 4096 ? n_big : 4096;
+      hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
+                         A);
+    }
+    hipLaunchKernelGGL(k_sum_sq_f64, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                       ctx->stream, nv, cinfo_deg.as<double>(), scalars.as<double>() + 1);
+    double exx = 0.0, a2x = 0.0;
+    MGX_TRY(read_scalar_f64(ctx, scalars.as<double>(), &exx));
+    MGX_TRY(read_scalar_f64(ctx, scalars.as<double>() + 1, &a2x));
+    curr_mod = exx * constant - a2x * constant * constant;
+    if ((curr_mod - prev_mod) < thresh) break;
+    prev_mod = curr_mod;
+    if (prev_mod < lower) prev_mod = lower;
+    hipLaunchKernelGGL(k_apply_updates, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                       ctx->stream, nv, cinfo_deg.as<double>(), cinfo_size.as<int32_t>(),
+                       cupd_deg.as<double>(), cupd_size.as<int32_t>());
+    // rotation (parallelLouvainMethod.cpp:268-274)
+    int32_t *tmp = p_past;
+    p_past = p_curr;
+    p_curr = p_target;
+    p_target = tmp;
+    if (iters > 100000) break;  // matches runMultiPhaseBasic's totItr cap scale
+  }
+
+  MGX_HIP_TRY(hipMemcpyAsync(C, p_past, nv * 4, hipMemcpyDeviceToDevice, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  *out_mod = curr_mod;
+  *iters_out = iters;
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+// renumberClustersContiguously: new id = rank of community's min vertex.
+mgx_status renumber(mgx_context *ctx, int32_t *C, int64_t nv, int64_t *n_clusters) {
+  DevBuf rep, keys_out, cand, cand_out, newid, nact;
+  MGX_HIP_TRY(rep.alloc(nv * 4));
+  hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, INT32_MAX, rep.as<int32_t>());
+  hipLaunchKernelGGL(k_rep_min, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, C, rep.as<int32_t>());
+  // sort (rep, c) ascending by rep; INT32_MAX reps (inactive ids) sort last.
+  MGX_HIP_TRY(cand.alloc(nv * 4));
+  MGX_HIP_TRY(cand_out.alloc(nv * 4));
+  MGX_HIP_TRY(keys_out.alloc(nv * 4));
+  hipLaunchKernelGGL(k_iota_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, cand.as<int32_t>());
+  size_t tmp_bytes = 0;
+  auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, (uint32_t *)rep.p,
+                                       (uint32_t *)keys_out.p, cand.as<int32_t>(),
+                                       cand_out.as<int32_t>(), nv, 0, 32, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::radix_sort_pairs(tmp, tmp_bytes, (uint32_t *)rep.p,
+                                  (uint32_t *)keys_out.p, cand.as<int32_t>(),
+                                  cand_out.as<int32_t>(), nv, 0, 32, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  // count active = reps != INT32_MAX: binary property; count via reduce on
+  // host copy of the boundary — simpler: count nonzero with a kernel.
+  DevBuf count;
+  MGX_HIP_TRY(count.alloc(8));
+  MGX_HIP_TRY(hipMemsetAsync(count.p, 0, 8, ctx->stream));
+  {
+    struct K {
+      static __global__ void count_active(int64_t n, const uint32_t *sorted_rep,
+                                          unsigned long long *out) {
+        unsigned long long acc = 0;
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+             i += (int64_t)gridDim.x * blockDim.x)
+          if (sorted_rep[i] != 0x7FFFFFFFu) ++acc;
+        __shared__ unsigned long long red[kBlock / 64];
+        for (int o = 32; o; o >>= 1) acc += __shfl_down(acc, o, 64);
+        if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+          unsigned long long s = 0;
+          for (int i = 0; i < kBlock / 64; ++i) s += red[i];
+          atomicAdd(out, s);
+        }
+      }
+    };
+    hipLaunchKernelGGL(K::count_active, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                       ctx->stream, nv, (uint32_t *)keys_out.p,
+                       (unsigned long long *)count.p);
+  }
+  unsigned long long n_active = 0;
+  MGX_HIP_TRY(hipMemcpyAsync(&n_active, count.p, 8, hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  MGX_HIP_TRY(newid.alloc(nv * 4));
+  hipLaunchKernelGGL(k_newid_from_sorted, dim3((uint32_t)grid_for((int64_t)n_active)),
+                     dim3(kBlock), 0, ctx->stream, (int64_t)n_active,
+                     cand_out.as<int32_t>(), newid.as<int32_t>());
+  hipLaunchKernelGGL(k_remap, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
+                     nv, C, newid.as<int32_t>());
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  *n_clusters = (int64_t)n_active;
+  return MGX_OK;
+}
+
+// buildNextLevelGraphOpt: aggregate (C[i] >= C[tail]) pairs, always-present
+// self loops, mirrored lower pairs.
+mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t n_clusters,
+                   Level *out) {
+  const int64_t ne2 = in.ne2;
+  DevBuf row_of_entry, keys, vals, keys_sorted, vals_sorted, u_keys, u_vals, u_count;
+  MGX_HIP_TRY(row_of_entry.alloc(ne2 * 4));
+  hipLaunchKernelGGL(k_row_of_entry, dim3((uint32_t)grid_for(in.nv)), dim3(kBlock), 0,
+                     ctx->stream, in.nv, in.row_ptr, row_of_entry.as<int32_t>());
+  MGX_HIP_TRY(keys.alloc(ne2 * 8));
+  MGX_HIP_TRY(vals.alloc(ne2 * 8));
+  const uint64_t sentinel = ~0ull;
+  hipLaunchKernelGGL(k_pair_keys, dim3((uint32_t)grid_for(ne2)), dim3(kBlock), 0,
+                     ctx->stream, ne2, in.col, in.row_ptr, in.nv, C, in.w,
+                     row_of_entry.as<int32_t>(), keys.as<uint64_t>(), vals.as<double>(),
+                     sentinel);
+  MGX_HIP_TRY(keys_sorted.alloc(ne2 * 8));
+  MGX_HIP_TRY(vals_sorted.alloc(ne2 * 8));
+  size_t tmp_bytes = 0;
+  auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys.as<uint64_t>(),
+                                       keys_sorted.as<uint64_t>(), vals.as<double>(),
+                                       vals_sorted.as<double>(), ne2, 0, 64, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::radix_sort_pairs(tmp, tmp_bytes, keys.as<uint64_t>(),
+                                  keys_sorted.as<uint64_t>(), vals.as<double>(),
+                                  vals_sorted.as<double>(), ne2, 0, 64, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  MGX_HIP_TRY(u_keys.alloc(ne2 * 8));
+  MGX_HIP_TRY(u_vals.alloc(ne2 * 8));
+  MGX_HIP_TRY(u_count.alloc(8));
+  err = rocprim::reduce_by_key(nullptr, tmp_bytes, keys_sorted.as<uint64_t>(),
+                               vals_sorted.as<double>(), ne2, u_keys.as<uint64_t>(),
+                               u_vals.as<double>(), (unsigned int *)u_count.p,
+                               rocprim::plus<double>(), rocprim::equal_to<uint64_t>(),
+                               ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::reduce_by_key(tmp, tmp_bytes, keys_sorted.as<uint64_t>(),
+                               vals_sorted.as<double>(), ne2, u_keys.as<uint64_t>(),
+                               u_vals.as<double>(), (unsigned int *)u_count.p,
+                               rocprim::plus<double>(), rocprim::equal_to<uint64_t>(),
+                               ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  unsigned int n_unique = 0;
+  MGX_HIP_TRY(hipMemcpyAsync(&n_unique, u_count.p, 4, hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  // drop the sentinel group if present (it sorts last)
+  int64_t n_pairs = n_unique;
+  if (n_pairs > 0) {
+    uint64_t last_key = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&last_key, u_keys.as<uint64_t>() + n_pairs - 1, 8,
+                               hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    if (last_key == sentinel) --n_pairs;
+  }
+
+  // CSR counts + self presence.
+  DevBuf counts, self_present;
+  MGX_HIP_TRY(counts.alloc(n_clusters * 4));
+  MGX_HIP_TRY(self_present.alloc(n_clusters * 4));
+  MGX_HIP_TRY(hipMemsetAsync(counts.p, 0, n_clusters * 4, ctx->stream));
+  MGX_HIP_TRY(hipMemsetAsync(self_present.p, 0, n_clusters * 4, ctx->stream));
+  if (n_pairs > 0)
+    hipLaunchKernelGGL(k_pair_counts, dim3((uint32_t)grid_for(n_pairs)), dim3(kBlock), 0,
+                       ctx->stream, n_pairs, u_keys.as<uint64_t>(), counts.as<uint32_t>(),
+                       self_present.as<uint32_t>());
+  hipLaunchKernelGGL(k_add_missing_self, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock),
+                     0, ctx->stream, n_clusters, self_present.as<uint32_t>(),
+                     counts.as<uint32_t>());
+
+  out->nv = n_clusters;
+  MGX_HIP_TRY(hipMalloc(&out->row_ptr, (n_clusters + 1) * 4));
+  // exclusive scan counts -> row_ptr
+  err = rocprim::exclusive_scan(nullptr, tmp_bytes, counts.as<uint32_t>(), out->row_ptr,
+                                0u, n_clusters, rocprim::plus<uint32_t>(), ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::exclusive_scan(tmp, tmp_bytes, counts.as<uint32_t>(), out->row_ptr, 0u,
+                                n_clusters, rocprim::plus<uint32_t>(), ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  uint32_t last_off = 0, last_cnt = 0;
+  MGX_HIP_TRY(hipMemcpyAsync(&last_off, out->row_ptr + n_clusters - 1, 4,
+                             hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipMemcpyAsync(&last_cnt, counts.as<uint32_t>() + n_clusters - 1, 4,
+                             hipMemcpyDeviceToHost, ctx->stream));
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  const uint32_t total = last_off + last_cnt;
+  MGX_HIP_TRY(hipMemcpyAsync(out->row_ptr + n_clusters, &total, 4, hipMemcpyHostToDevice,
+                             ctx->stream));
+  out->ne2 = total;
+  MGX_HIP_TRY(hipMalloc(&out->col, (total > 0 ? total : 1) * 4));
+  MGX_HIP_TRY(hipMalloc(&out->w, (total > 0 ? total : 1) * 8));
+  // cursor = row_ptr copy (reuse counts buffer)
+  MGX_HIP_TRY(hipMemcpyAsync(counts.p, out->row_ptr, n_clusters * 4,
+                             hipMemcpyDeviceToDevice, ctx->stream));
+  if (n_pairs > 0)
+    hipLaunchKernelGGL(k_pair_scatter, dim3((uint32_t)grid_for(n_pairs)), dim3(kBlock), 0,
+                       ctx->stream, n_pairs, u_keys.as<uint64_t>(), u_vals.as<double>(),
+                       counts.as<uint32_t>(), out->col, out->w);
+  hipLaunchKernelGGL(k_self_scatter, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock), 0,
+                     ctx->stream, n_clusters, self_present.as<uint32_t>(),
+                     counts.as<uint32_t>(), out->col, out->w);
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
+}
+
+void free_level(Level *L, bool own) {
+  if (!own) return;
+  if (L->row_ptr) (void)hipFree(L->row_ptr);
+  if (L->col) (void)hipFree(L->col);
+  if (L->w) (void)hipFree(L->w);
+  L->row_ptr = nullptr;
+  L->col = nullptr;
+  L->w = nullptr;
+}
+
+__global__ void k_compose(int64_t nv, int32_t *c_orig, const int32_t *C) {
+  // runMultiPhaseBasic.cpp:101-106: C_orig[i] = C[C_orig[i]] when >= 0.
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nv;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (c_orig[i] >= 0) c_orig[i] = C[c_orig[i]];
+  }
+}
+
+__global__ void k_widen_i32_to_i64(int64_t n, const int32_t *in, int64_t *out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int64_t)in[i];
+}
+
+}  // namespace
+
 mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
                             int64_t *out_community, int64_t *n_communities) {
-  (void)ctx;
-  (void)g;
-  (void)threshold;
-  (void)out_community;
-  (void)n_communities;
-  mgx_set_error("mgx_louvain: GPU implementation pending (round 1 WIP)");
-  return MGX_ERR_NOT_SUPPORTED;
+  if (!(g->flags & MGX_BUILD_SYM_CSR)) {
+    mgx_set_error("louvain needs a graph built with MGX_BUILD_SYM_CSR");
+    return MGX_ERR_INVALID_ARGUMENT;
+  }
+  const int64_t nv0 = g->n_vertices;
+  if (n_communities) *n_communities = 0;
+  if (out_community) {
+    for (int64_t i = 0; i < nv0; ++i) out_community[i] = -1;
+  }
+  if (nv0 == 0 || g->n_edges == 0) return MGX_OK;
+
+  // Level 0: widen weights to fp64 (unweighted graphs: all 1.0).
+  Level L;
+  L.nv = nv0;
+  L.row_ptr = g->sym_row_ptr;
+  L.col = g->sym_col;
+  L.ne2 = 2 * g->n_edges;
+  DevBuf w0;
+  MGX_HIP_TRY(w0.alloc(L.ne2 * 8));
+  if (g->sym_w) {
+    hipLaunchKernelGGL(k_f32_to_f64, dim3((uint32_t)grid_for(L.ne2)), dim3(kBlock), 0,
+                       ctx->stream, L.ne2, g->sym_w, (double *)w0.p);
+  } else {
+    struct K {
+      static __global__ void fill1(int64_t n, double *p) {
+        for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+             i += (int64_t)gridDim.x * blockDim.x)
+          p[i] = 1.0;
+      }
+    };
+    hipLaunchKernelGGL(K::fill1, dim3((uint32_t)grid_for(L.ne2)), dim3(kBlock), 0,
+                       ctx->stream, L.ne2, (double *)w0.p);
+  }
+  L.w = (double *)w0.p;
+  bool own_level = false;
+
+  DevBuf c_orig_d, c_level;
+  MGX_HIP_TRY(c_orig_d.alloc(nv0 * 4));
+  hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for(nv0)), dim3(kBlock), 0,
+                     ctx->stream, nv0, -1, c_orig_d.as<int32_t>());
+
+  double prev_mod = -1.0, curr_mod = -1.0;
+  int64_t phase = 1, tot_itr = 0;
+  mgx_status status = MGX_OK;
+  while (true) {
+    DevBuf C;
+    if (C.alloc(L.nv * 4) != hipSuccess) {
+      status = MGX_ERR_OUT_OF_MEMORY;
+      break;
+    }
+    prev_mod = curr_mod;
+    int64_t iters = 0;
+    status = louvain_level(ctx, L, prev_mod, threshold, C.as<int32_t>(), &curr_mod, &iters);
+    if (status != MGX_OK) break;
+    tot_itr += iters;
+    int64_t n_clusters = 0;
+    status = renumber(ctx, C.as<int32_t>(), L.nv, &n_clusters);
+    if (status != MGX_OK) break;
+    if (phase == 1) {
+      MGX_HIP_TRY(hipMemcpyAsync(c_orig_d.p, C.p, nv0 * 4, hipMemcpyDeviceToDevice,
+                                 ctx->stream));
+    } else {
+      hipLaunchKernelGGL(k_compose, dim3((uint32_t)grid_for(nv0)), dim3(kBlock), 0,
+                         ctx->stream, nv0, c_orig_d.as<int32_t>(), C.as<int32_t>());
+    }
+    if (phase > 200 || tot_itr > 100000) break;
+    if ((curr_mod - prev_mod) > threshold) {
+      Level next;
+      status = coarsen(ctx, L, C.as<int32_t>(), n_clusters, &next);
+      free_level(&L, own_level);
+      if (status != MGX_OK) break;
+      L = next;
+      own_level = true;
+      ++phase;
+    } else {
+      break;
+    }
+  }
+  free_level(&L, own_level);
+  if (status != MGX_OK) return status;
+
+  // Download.
+  DevBuf wide;
+  MGX_HIP_TRY(wide.alloc(nv0 * 8));
+  hipLaunchKernelGGL(k_widen_i32_to_i64, dim3((uint32_t)grid_for(nv0)), dim3(kBlock), 0,
+                     ctx->stream, nv0, c_orig_d.as<int32_t>(), wide.as<int64_t>());
+  if (out_community) {
+    MGX_HIP_TRY(hipMemcpyAsync(out_community, wide.p, nv0 * 8, hipMemcpyDeviceToHost,
+                               ctx->stream));
+  }
+  MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  if (n_communities && out_community) {
+    int64_t mx = -1;
+    for (int64_t i = 0; i < nv0; ++i)
+      if (out_community[i] > mx) mx = out_community[i];
+    *n_communities = mx + 1;
+  }
+  MGX_HIP_TRY(hipGetLastError());
+  return MGX_OK;
 }
