@@ -138,9 +138,77 @@ __global__ void k_bin_keys(int64_t rows, const uint32_t *row_ptr, uint32_t *keys
 }
 
 __global__ void k_count_bins(int64_t rows, const uint32_t *keys, uint32_t *counts4) {
+  // Per-block LDS aggregation first: a naive 4-counter global histogram
+  // measured 706 ms at RMAT-26 (atomic hotspot); this form is ~ms.
+  __shared__ uint32_t local[4];
+  if (threadIdx.x < 4) local[threadIdx.x] = 0;
+  __syncthreads();
   for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
        r += (int64_t)gridDim.x * blockDim.x)
-    atomicAdd(&counts4[keys[r]], 1u);
+    atomicAdd(&local[keys[r]], 1u);
+  __syncthreads();
+  if (threadIdx.x < 4 && local[threadIdx.x]) atomicAdd(&counts4[threadIdx.x], local[threadIdx.x]);
+}
+
+__global__ void k_pack_pairs(int64_t n_edges, const int32_t *src, const int32_t *dst,
+                             uint64_t *keys) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x)
+    keys[i] = ((uint64_t)(uint32_t)dst[i] << 32) | (uint32_t)src[i];
+}
+
+__global__ void k_pack_pairs_ranged(int64_t n_edges, const int32_t *src, const int32_t *dst,
+                                    int32_t lo, int32_t hi, uint64_t *keys) {
+  // Out-of-range edges get the max key: they sort last and are trimmed.
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t d = dst[i];
+    keys[i] = (d >= lo && d < hi)
+                  ? ((uint64_t)(uint32_t)(d - lo) << 32) | (uint32_t)src[i]
+                  : ~0ull;
+  }
+}
+
+__global__ void k_unpack_cols(int64_t n, const uint64_t *keys, int32_t *col) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    col[i] = (int32_t)(uint32_t)keys[i];
+}
+
+// Build a sorted in-CSR column array: cols within each row in ascending
+// order (gather locality: adjacent cols share cache lines — the measured
+// unsorted-scatter build over-fetched 3x algorithmic bytes in the sweep).
+mgx_status build_sorted_cols(mgx_context *ctx, const int32_t *d_src, const int32_t *d_dst,
+                             int64_t n_edges, int64_t key_rows, bool ranged, int32_t lo,
+                             int32_t hi, int32_t *col, int64_t col_count) {
+  if (n_edges == 0) return MGX_OK;
+  uint64_t *keys = nullptr, *keys_out = nullptr;
+  MGX_HIP_TRY(hipMalloc(&keys, n_edges * sizeof(uint64_t)));
+  MGX_HIP_TRY(hipMalloc(&keys_out, n_edges * sizeof(uint64_t)));
+  if (ranged) {
+    hipLaunchKernelGGL(k_pack_pairs_ranged, dim3(grid_for(n_edges)), dim3(kBlock), 0,
+                       ctx->stream, n_edges, d_src, d_dst, lo, hi, keys);
+  } else {
+    hipLaunchKernelGGL(k_pack_pairs, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                       n_edges, d_src, d_dst, keys);
+  }
+  int end_bit = 33;
+  while ((1ll << (end_bit - 32)) < key_rows + 1) ++end_bit;
+  if (ranged) end_bit = 64;  // the ~0 sentinel must sort last
+  size_t tmp_bytes = 0;
+  auto err = rocprim::radix_sort_keys(nullptr, tmp_bytes, keys, keys_out, n_edges, 0,
+                                      end_bit, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::radix_sort_keys(tmp, tmp_bytes, keys, keys_out, n_edges, 0, end_bit,
+                                 ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  hipLaunchKernelGGL(k_unpack_cols, dim3(grid_for(col_count)), dim3(kBlock), 0, ctx->stream,
+                     col_count, keys_out, col);
+  MGX_HIP_TRY(hipFree(keys));
+  MGX_HIP_TRY(hipFree(keys_out));
+  return MGX_OK;
 }
 
 __global__ void k_i64_to_i32(int64_t n, const int64_t *in, int32_t *out) {
@@ -302,11 +370,7 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     MGX_HIP_TRY(hipMalloc(&g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
     MGX_TRY(scan_counts(ctx, counts, V, g->in_row_ptr));
     MGX_HIP_TRY(hipMalloc(&g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
-    // cursor = copy of row_ptr[0..V)
-    MGX_HIP_TRY(hipMemcpyAsync(counts, g->in_row_ptr, V * sizeof(uint32_t),
-                               hipMemcpyDeviceToDevice, ctx->stream));
-    hipLaunchKernelGGL(k_scatter_in, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
-                       d_src, d_dst, counts, g->in_col);
+    MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, E, V, false, 0, 0, g->in_col, E));
     MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, V, &g->bins_in));
   }
 
@@ -378,11 +442,8 @@ mgx_status mgx_build_sharded_in_csr(mgx_context *ctx, const int32_t *d_src,
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   g->in_edges = local_edges;
   MGX_HIP_TRY(hipMalloc(&g->in_col, (local_edges > 0 ? local_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(hipMemcpyAsync(counts, g->in_row_ptr, rows * sizeof(uint32_t),
-                             hipMemcpyDeviceToDevice, ctx->stream));
-  hipLaunchKernelGGL(k_scatter_in_ranged, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream,
-                     E, d_src, d_dst, (int32_t)row_begin, (int32_t)row_end, counts,
-                     g->in_col);
+  MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, E, rows, true, (int32_t)row_begin,
+                            (int32_t)row_end, g->in_col, local_edges));
   MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, rows, &g->bins_in));
   MGX_HIP_TRY(hipFree(counts));
 

@@ -54,7 +54,32 @@ __device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) 
     if (ri < nrows) {
       row = rows_list[ri];
       const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
-      for (uint32_t j = s + sub; j < e; j += LANES) acc += (double)A.contrib_old[A.col[j]];
+      if constexpr (LANES >= 64) {
+        // Wide rows: scalar head to 16-B alignment, then int4 nontemporal
+        // column loads (the col stream is read exactly once — keep it out
+        // of L1 so the contrib gathers stay cached; G13/nt-weights) with 4
+        // independent gathers per lane per iteration for latency hiding.
+        uint32_t s_al = (s + 3u) & ~3u;
+        if (s_al > e) s_al = e;
+        for (uint32_t j = s + sub; j < s_al; j += LANES)
+          acc += (double)A.contrib_old[A.col[j]];
+        const uint32_t nvec = (e - s_al) / 4;
+        const int4 *col4 = reinterpret_cast<const int4 *>(A.col + s_al);
+        for (uint32_t c = sub; c < nvec; c += LANES) {
+          const int4 cc = __builtin_nontemporal_load(col4 + c);
+          acc += (double)A.contrib_old[cc.x];
+          acc += (double)A.contrib_old[cc.y];
+          acc += (double)A.contrib_old[cc.z];
+          acc += (double)A.contrib_old[cc.w];
+        }
+        for (uint32_t j = s_al + nvec * 4 + sub; j < e; j += LANES)
+          acc += (double)A.contrib_old[A.col[j]];
+      } else {
+        // Short rows: a LANES-wide group reads consecutive cols (16/64-B
+        // granules); nt keeps the one-pass col stream out of L1.
+        for (uint32_t j = s + sub; j < e; j += LANES)
+          acc += (double)A.contrib_old[__builtin_nontemporal_load(A.col + j)];
+      }
     }
     if constexpr (LANES <= 64) {
       for (int o = LANES / 2; o; o >>= 1) acc += __shfl_down(acc, o, LANES);
