@@ -64,9 +64,10 @@ __device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) 
         for (uint32_t j = s + sub; j < s_al; j += LANES)
           acc += (double)A.contrib_old[A.col[j]];
         const uint32_t nvec = (e - s_al) / 4;
-        const int4 *col4 = reinterpret_cast<const int4 *>(A.col + s_al);
+        typedef int v4i __attribute__((ext_vector_type(4)));
+        const v4i *col4 = reinterpret_cast<const v4i *>(A.col + s_al);
         for (uint32_t c = sub; c < nvec; c += LANES) {
-          const int4 cc = __builtin_nontemporal_load(col4 + c);
+          const v4i cc = __builtin_nontemporal_load(col4 + c);
           acc += (double)A.contrib_old[cc.x];
           acc += (double)A.contrib_old[cc.y];
           acc += (double)A.contrib_old[cc.z];
