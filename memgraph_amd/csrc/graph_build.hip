@@ -126,28 +126,30 @@ __global__ void k_inv_outdeg(int64_t n, const uint32_t *deg, float *inv) {
     inv[i] = deg[i] ? 1.0f / (float)deg[i] : 0.0f;
 }
 
-__global__ void k_bin_keys(int64_t rows, const uint32_t *row_ptr, uint32_t *keys,
-                           uint32_t *vals) {
+__global__ void k_bin_keys(int64_t rows, const uint32_t *lo, const uint32_t *hi,
+                           int include_zero, uint32_t *keys, uint32_t *vals) {
   for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
        r += (int64_t)gridDim.x * blockDim.x) {
-    const uint32_t deg = row_ptr[r + 1] - row_ptr[r];
+    const uint32_t deg = hi[r] - lo[r];
     uint32_t k = deg < 8 ? 0u : deg < 64 ? 1u : deg < 1024 ? 2u : 3u;
+    if (deg == 0 && !include_zero) k = 4u;  // dropped (sorts last)
     keys[r] = k;
     vals[r] = (uint32_t)r;
   }
 }
 
-__global__ void k_count_bins(int64_t rows, const uint32_t *keys, uint32_t *counts4) {
-  // Per-block LDS aggregation first: a naive 4-counter global histogram
-  // measured 706 ms at RMAT-26 (atomic hotspot); this form is ~ms.
-  __shared__ uint32_t local[4];
-  if (threadIdx.x < 4) local[threadIdx.x] = 0;
+__global__ void k_count_bins(int64_t rows, const uint32_t *keys, uint32_t *counts5) {
+  // Per-block LDS aggregation first: a naive global histogram measured
+  // 706 ms at RMAT-26 (atomic hotspot); this form is ~ms. Key 4 counts the
+  // dropped zero-degree rows.
+  __shared__ uint32_t local[5];
+  if (threadIdx.x < 5) local[threadIdx.x] = 0;
   __syncthreads();
   for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
        r += (int64_t)gridDim.x * blockDim.x)
     atomicAdd(&local[keys[r]], 1u);
   __syncthreads();
-  if (threadIdx.x < 4 && local[threadIdx.x]) atomicAdd(&counts4[threadIdx.x], local[threadIdx.x]);
+  if (threadIdx.x < 5 && local[threadIdx.x]) atomicAdd(&counts5[threadIdx.x], local[threadIdx.x]);
 }
 
 __global__ void k_pack_pairs(int64_t n_edges, const int32_t *src, const int32_t *dst,
@@ -208,6 +210,74 @@ mgx_status build_sorted_cols(mgx_context *ctx, const int32_t *d_src, const int32
                      col_count, keys_out, col);
   MGX_HIP_TRY(hipFree(keys));
   MGX_HIP_TRY(hipFree(keys_out));
+  return MGX_OK;
+}
+
+__global__ void k_pack_sym(int64_t n_edges, const int32_t *src, const int32_t *dst,
+                           const float *w, uint64_t *keys, float *vals) {
+  // Each input edge twice: (s->d) and (d->s), the GetGrappoloSuitableGraph
+  // layout (louvain.cpp:176-233), as sortable (row<<32|col) keys.
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t a = (uint32_t)src[i], b = (uint32_t)dst[i];
+    const float wi = w ? w[i] : 1.0f;
+    keys[2 * i] = ((uint64_t)a << 32) | b;
+    keys[2 * i + 1] = ((uint64_t)b << 32) | a;
+    if (vals) {
+      vals[2 * i] = wi;
+      vals[2 * i + 1] = wi;
+    }
+  }
+}
+
+// Sorted symmetric CSR (cols ascending within each row) via pair sort.
+mgx_status build_sorted_sym(mgx_context *ctx, const int32_t *d_src, const int32_t *d_dst,
+                            const float *d_w, int64_t n_edges, int64_t n_vertices,
+                            int32_t *col, float *out_w) {
+  if (n_edges == 0) return MGX_OK;
+  const int64_t n2 = 2 * n_edges;
+  uint64_t *keys = nullptr, *keys_out = nullptr;
+  float *vals = nullptr, *vals_out = nullptr;
+  MGX_HIP_TRY(hipMalloc(&keys, n2 * sizeof(uint64_t)));
+  MGX_HIP_TRY(hipMalloc(&keys_out, n2 * sizeof(uint64_t)));
+  if (out_w) {
+    MGX_HIP_TRY(hipMalloc(&vals, n2 * sizeof(float)));
+    MGX_HIP_TRY(hipMalloc(&vals_out, n2 * sizeof(float)));
+  }
+  hipLaunchKernelGGL(k_pack_sym, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                     n_edges, d_src, d_dst, d_w, keys, vals);
+  int end_bit = 33;
+  while ((1ll << (end_bit - 32)) < n_vertices + 1) ++end_bit;
+  size_t tmp_bytes = 0;
+  hipError_t err;
+  if (out_w) {
+    err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys, keys_out, vals, vals_out, n2,
+                                    0, end_bit, ctx->stream);
+  } else {
+    err = rocprim::radix_sort_keys(nullptr, tmp_bytes, keys, keys_out, n2, 0, end_bit,
+                                   ctx->stream);
+  }
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  if (out_w) {
+    err = rocprim::radix_sort_pairs(tmp, tmp_bytes, keys, keys_out, vals, vals_out, n2, 0,
+                                    end_bit, ctx->stream);
+  } else {
+    err = rocprim::radix_sort_keys(tmp, tmp_bytes, keys, keys_out, n2, 0, end_bit,
+                                   ctx->stream);
+  }
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  hipLaunchKernelGGL(k_unpack_cols, dim3(grid_for(n2)), dim3(kBlock), 0, ctx->stream, n2,
+                     keys_out, col);
+  if (out_w) {
+    MGX_HIP_TRY(hipMemcpyAsync(out_w, vals_out, n2 * sizeof(float),
+                               hipMemcpyDeviceToDevice, ctx->stream));
+  }
+  MGX_HIP_TRY(hipFree(keys));
+  MGX_HIP_TRY(hipFree(keys_out));
+  if (vals) MGX_HIP_TRY(hipFree(vals));
+  if (vals_out) MGX_HIP_TRY(hipFree(vals_out));
   return MGX_OK;
 }
 
@@ -282,8 +352,8 @@ mgx_status mgx_gen_weights_device(mgx_context *ctx, int64_t n_edges, uint64_t se
   return MGX_OK;
 }
 
-mgx_status mgx_build_bins(mgx_context *ctx, const uint32_t *row_ptr, int64_t rows,
-                          mgx_bins *bins) {
+mgx_status mgx_build_bins_range(mgx_context *ctx, const uint32_t *lo, const uint32_t *hi,
+                                int64_t rows, bool include_zero, mgx_bins *bins) {
   if (rows == 0) {
     bins->rows = nullptr;
     return MGX_OK;
@@ -293,28 +363,29 @@ mgx_status mgx_build_bins(mgx_context *ctx, const uint32_t *row_ptr, int64_t row
   MGX_HIP_TRY(hipMalloc(&keys, rows * sizeof(uint32_t)));
   MGX_HIP_TRY(hipMalloc(&keys_out, rows * sizeof(uint32_t)));
   MGX_HIP_TRY(hipMalloc(&vals, rows * sizeof(uint32_t)));
-  uint32_t *counts4 = nullptr;
-  MGX_HIP_TRY(hipMalloc(&counts4, 4 * sizeof(uint32_t)));
-  MGX_HIP_TRY(hipMemsetAsync(counts4, 0, 16, ctx->stream));
+  uint32_t *counts5 = nullptr;
+  MGX_HIP_TRY(hipMalloc(&counts5, 5 * sizeof(uint32_t)));
+  MGX_HIP_TRY(hipMemsetAsync(counts5, 0, 20, ctx->stream));
 
   hipLaunchKernelGGL(k_bin_keys, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream, rows,
-                     row_ptr, keys, vals);
+                     lo, hi, include_zero ? 1 : 0, keys, vals);
   hipLaunchKernelGGL(k_count_bins, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream, rows,
-                     keys, counts4);
+                     keys, counts5);
 
-  // Stable 2-bit radix sort: within a bin, rows stay in ascending id order.
+  // Stable 3-bit radix sort: within a bin, rows stay in ascending id order;
+  // key 4 (dropped rows) lands past the used prefix.
   size_t tmp_bytes = 0;
   auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys, keys_out, vals,
-                                       (uint32_t *)bins->rows, rows, 0, 2, ctx->stream);
+                                       (uint32_t *)bins->rows, rows, 0, 3, ctx->stream);
   if (err != hipSuccess) return MGX_ERR_HIP;
   void *tmp = nullptr;
   MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
   err = rocprim::radix_sort_pairs(tmp, tmp_bytes, keys, keys_out, vals,
-                                  (uint32_t *)bins->rows, rows, 0, 2, ctx->stream);
+                                  (uint32_t *)bins->rows, rows, 0, 3, ctx->stream);
   if (err != hipSuccess) return MGX_ERR_HIP;
 
-  uint32_t h_counts[4] = {0, 0, 0, 0};
-  MGX_HIP_TRY(hipMemcpyAsync(h_counts, counts4, 16, hipMemcpyDeviceToHost, ctx->stream));
+  uint32_t h_counts[5] = {0, 0, 0, 0, 0};
+  MGX_HIP_TRY(hipMemcpyAsync(h_counts, counts5, 20, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   for (int b = 0; b < 4; ++b) bins->count[b] = h_counts[b];
 
@@ -330,7 +401,80 @@ mgx_status mgx_build_bins(mgx_context *ctx, const uint32_t *row_ptr, int64_t row
   MGX_HIP_TRY(hipFree(keys));
   MGX_HIP_TRY(hipFree(keys_out));
   MGX_HIP_TRY(hipFree(vals));
-  MGX_HIP_TRY(hipFree(counts4));
+  MGX_HIP_TRY(hipFree(counts5));
+  return MGX_OK;
+}
+
+mgx_status mgx_build_bins(mgx_context *ctx, const uint32_t *row_ptr, int64_t rows,
+                          mgx_bins *bins) {
+  return mgx_build_bins_range(ctx, row_ptr, row_ptr + 1, rows, /*include_zero=*/true,
+                              bins);
+}
+
+namespace {
+// Binary-search the sorted cols of each row for the stripe boundary value.
+__global__ void k_stripe_search(int64_t rows, const uint32_t *row_ptr, const int32_t *col,
+                                int32_t boundary, uint32_t *out) {
+  for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
+       r += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t lo = row_ptr[r], hi = row_ptr[r + 1];
+    while (lo < hi) {
+      uint32_t mid = (lo + hi) / 2;
+      if (col[mid] < boundary) lo = mid + 1;
+      else hi = mid;
+    }
+    out[r] = lo;
+  }
+}
+}  // namespace
+
+mgx_status mgx_build_stripes(mgx_context *ctx, mgx_graph *g) {
+  // Called while row_end is still the clamped row count (the sharded path
+  // re-pads it afterwards), so this is exactly the in-CSR row count.
+  const int64_t rows = g->row_end - g->row_begin;
+  const int64_t V = g->n_vertices;
+  int n_stripes = 1;
+  const char *env = getenv("MGX_PR_STRIPES");
+  if (env && atoi(env) > 0) {
+    n_stripes = atoi(env);
+  } else {
+    // auto: keep the gathered contrib stripe (f32[V/S]) under ~96 MB so it
+    // stays Infinity-Cache-resident beside the streaming col traffic.
+    const int64_t contrib_bytes = V * 4;
+    n_stripes = (int)((contrib_bytes + (96 << 20) - 1) / (96 << 20));
+    if (n_stripes < 1) n_stripes = 1;
+  }
+  if (n_stripes > 8) n_stripes = 8;
+  g->n_stripes = n_stripes;
+  if (n_stripes == 1) return MGX_OK;
+  g->stripe_width = (V + n_stripes - 1) / n_stripes;
+
+  MGX_HIP_TRY(hipMalloc(&g->stripe_ptr, (size_t)(n_stripes + 1) * rows * sizeof(uint32_t)));
+  // boundaries: sp[0] = row starts, sp[S] = row ends, sp[s] = searchsorted.
+  struct CopyK {
+    static __global__ void shift(int64_t rows, const uint32_t *row_ptr, uint32_t *lo,
+                                 uint32_t *hi) {
+      for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < rows;
+           r += (int64_t)gridDim.x * blockDim.x) {
+        lo[r] = row_ptr[r];
+        hi[r] = row_ptr[r + 1];
+      }
+    }
+  };
+  hipLaunchKernelGGL(CopyK::shift, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream, rows,
+                     g->in_row_ptr, g->stripe_ptr, g->stripe_ptr + (size_t)n_stripes * rows);
+  for (int sIdx = 1; sIdx < n_stripes; ++sIdx) {
+    hipLaunchKernelGGL(k_stripe_search, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream,
+                       rows, g->in_row_ptr, g->in_col,
+                       (int32_t)((int64_t)sIdx * g->stripe_width),
+                       g->stripe_ptr + (size_t)sIdx * rows);
+  }
+  for (int sIdx = 0; sIdx < n_stripes; ++sIdx) {
+    const bool include_zero = (sIdx == 0) || (sIdx == n_stripes - 1);
+    MGX_TRY(mgx_build_bins_range(ctx, g->stripe_ptr + (size_t)sIdx * rows,
+                                 g->stripe_ptr + (size_t)(sIdx + 1) * rows, rows,
+                                 include_zero, &g->stripe_bins[sIdx]));
+  }
   return MGX_OK;
 }
 
@@ -372,6 +516,7 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     MGX_HIP_TRY(hipMalloc(&g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
     MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, E, V, false, 0, 0, g->in_col, E));
     MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, V, &g->bins_in));
+    MGX_TRY(mgx_build_stripes(ctx, g));
   }
 
   if (flags & MGX_BUILD_SYM_CSR) {
@@ -384,10 +529,7 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     if (flags & MGX_BUILD_WEIGHTED) {
       MGX_HIP_TRY(hipMalloc(&g->sym_w, (E > 0 ? 2 * E : 1) * sizeof(float)));
     }
-    MGX_HIP_TRY(hipMemcpyAsync(counts, g->sym_row_ptr, V * sizeof(uint32_t),
-                               hipMemcpyDeviceToDevice, ctx->stream));
-    hipLaunchKernelGGL(k_scatter_sym, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
-                       d_src, d_dst, d_w, counts, g->sym_col, g->sym_w);
+    MGX_TRY(build_sorted_sym(ctx, d_src, d_dst, d_w, E, V, g->sym_col, g->sym_w));
     MGX_TRY(mgx_build_bins(ctx, g->sym_row_ptr, V, &g->bins_sym));
   }
 
@@ -445,6 +587,7 @@ mgx_status mgx_build_sharded_in_csr(mgx_context *ctx, const int32_t *d_src,
   MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, E, rows, true, (int32_t)row_begin,
                             (int32_t)row_end, g->in_col, local_edges));
   MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, rows, &g->bins_in));
+  MGX_TRY(mgx_build_stripes(ctx, g));
   MGX_HIP_TRY(hipFree(counts));
 
   MGX_HIP_TRY(hipEventRecord(ev1, ctx->stream));

@@ -83,6 +83,16 @@ struct mgx_graph {
   mgx_bins bins_in;   // over in-CSR rows
   mgx_bins bins_sym;  // over sym-CSR rows
 
+  // Source-striped in-CSR view (PageRank at large V): cols are sorted
+  // within each row, so stripe s of row v is the contiguous sub-range
+  // [stripe_ptr[s*rows + v], stripe_ptr[(s+1)*rows + v]) whose sources lie
+  // in [s*stripe_width, (s+1)*stripe_width) — the gathered contrib stripe
+  // then fits in the Infinity Cache. n_stripes == 1 => unstriped.
+  int n_stripes = 1;
+  int64_t stripe_width = 0;
+  uint32_t *stripe_ptr = nullptr;          // [(n_stripes+1) * rows]
+  mgx_bins stripe_bins[8];                 // per stripe (max 8 stripes)
+
   double build_ms = 0.0;
 };
 
@@ -99,6 +109,13 @@ mgx_status mgx_gen_weights_device(mgx_context *ctx, int64_t n_edges, uint64_t se
 // Build the degree-bin work lists for a CSR with `rows` rows.
 mgx_status mgx_build_bins(mgx_context *ctx, const uint32_t *row_ptr, int64_t rows,
                           mgx_bins *bins);
+// Degree-bin work lists from explicit per-row [lo, hi) ranges; rows with
+// zero degree are dropped unless include_zero.
+mgx_status mgx_build_bins_range(mgx_context *ctx, const uint32_t *lo, const uint32_t *hi,
+                                int64_t rows, bool include_zero, mgx_bins *bins);
+// Source-stripe the (sorted) in-CSR of g; n_stripes chosen from V (env
+// MGX_PR_STRIPES overrides; 1 = disabled).
+mgx_status mgx_build_stripes(mgx_context *ctx, mgx_graph *g);
 
 // Sharded in-CSR build: keeps only edges with dst in [row_begin,row_end);
 // out_degree stays global.
@@ -117,6 +134,7 @@ struct mgx_pagerank_run {
   int cur = 0;
   uint32_t *d_delta = nullptr;  // Linf as ordered-uint f32
   double *d_scratch = nullptr;  // sum + f64 output [V+1]
+  double *d_partial = nullptr;  // striped sweep: fp64 row partials [rows]
   int64_t iterations = 0;
   // HIP-event timing of the sweep kernel (the dominant kernel).
   std::vector<hipEvent_t> ev_start, ev_stop;

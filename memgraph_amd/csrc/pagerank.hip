@@ -37,6 +37,14 @@ struct PrArgs {
   float base_term;          // (1-d)/N
   float damping;
   uint32_t *delta_max;      // f32-as-ordered-uint
+  // Source-striped sweep (large V): per-stripe row sub-ranges
+  // and fp64 row partials. mode: 0 = direct (single launch),
+  // 1 = first stripe (partial = acc), 2 = middle (partial += acc),
+  // 3 = last stripe (finish from partial + acc).
+  const uint32_t *sp_lo;
+  const uint32_t *sp_hi;
+  double *partial;
+  int mode;
 };
 
 template <int LANES>
@@ -53,7 +61,7 @@ __device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) 
     int32_t row = -1;
     if (ri < nrows) {
       row = rows_list[ri];
-      const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
+      const uint32_t s = A.sp_lo[row], e = A.sp_hi[row];
       if constexpr (LANES >= 64) {
         // Wide rows: scalar head to 16-B alignment, then int4 nontemporal
         // column loads (the col stream is read exactly once — keep it out
@@ -91,13 +99,20 @@ __device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) 
       if (threadIdx.x == 0) acc = red[0] + red[1] + red[2] + red[3];
     }
     if (sub == 0 && row >= 0 && (LANES <= 64 || threadIdx.x == 0)) {
-      const float newr = (float)((double)A.base_term + (double)A.damping * acc);
-      const int64_t gv = A.row_base + row;
-      const float oldr = A.rank_old[gv];
-      A.rank_new[gv] = newr;
-      A.contrib_new[gv] = newr * A.inv_outdeg[gv];
-      const float d = fabsf(newr - oldr);
-      if (d > maxd) maxd = d;
+      if (A.mode == 1) {
+        A.partial[row] = acc;
+      } else if (A.mode == 2) {
+        A.partial[row] += acc;
+      } else {
+        if (A.mode == 3) acc += A.partial[row];
+        const float newr = (float)((double)A.base_term + (double)A.damping * acc);
+        const int64_t gv = A.row_base + row;
+        const float oldr = A.rank_old[gv];
+        A.rank_new[gv] = newr;
+        A.contrib_new[gv] = newr * A.inv_outdeg[gv];
+        const float d = fabsf(newr - oldr);
+        if (d > maxd) maxd = d;
+      }
     }
     if constexpr (LANES > 64) __syncthreads();
   }
@@ -180,16 +195,6 @@ mgx_status queue_one_iteration(mgx_pagerank_run *run, bool track_delta) {
   PrArgs A;
   A.row_ptr = g->in_row_ptr;
   A.col = g->in_col;
-  A.bin_rows = g->bins_in.rows;
-  int64_t off = 0, goff = 0;
-  for (int b = 0; b < 4; ++b) {
-    A.n[b] = g->bins_in.count[b];
-    A.off[b] = off;
-    off += A.n[b];
-    A.goff[b] = goff;
-    A.grid[b] = g->bins_in.grid[b];
-    goff += A.grid[b];
-  }
   A.contrib_old = run->contrib[run->cur];
   A.rank_old = run->rank[run->cur];
   A.rank_new = run->rank[1 - run->cur];
@@ -199,12 +204,16 @@ mgx_status queue_one_iteration(mgx_pagerank_run *run, bool track_delta) {
   A.base_term = (float)((1.0 - run->damping) / (double)V);
   A.damping = (float)run->damping;
   A.delta_max = track_delta ? run->d_delta : nullptr;
+  A.partial = run->d_partial;
 
   if (track_delta) MGX_HIP_TRY(hipMemsetAsync(run->d_delta, 0, 4, ctx->stream));
 
-  const int64_t total_grid = goff;
-  if (total_grid > 0 && V > 0) {
-    // Event-bracket the sweep (the dominant kernel): feeds roofline.achieved.
+  const int n_stripes = g->n_stripes;
+  // in-CSR row count (row_end may be re-padded for the allgather shard).
+  const int64_t rows = (g->row_end < V ? g->row_end : V) - g->row_begin;
+  if (V > 0) {
+    // Event-bracket the whole per-iteration sweep group (1 launch when
+    // unstriped, n_stripes launches otherwise): feeds roofline.achieved.
     if (run->ev_used >= (int64_t)run->ev_start.size()) {
       hipEvent_t e0, e1;
       MGX_HIP_TRY(hipEventCreate(&e0));
@@ -213,8 +222,32 @@ mgx_status queue_one_iteration(mgx_pagerank_run *run, bool track_delta) {
       run->ev_stop.push_back(e1);
     }
     MGX_HIP_TRY(hipEventRecord(run->ev_start[run->ev_used], ctx->stream));
-    hipLaunchKernelGGL(k_pr_sweep, dim3((uint32_t)total_grid), dim3(kBlock), 0, ctx->stream,
-                       A);
+    for (int sIdx = 0; sIdx < n_stripes; ++sIdx) {
+      const mgx_bins &bins = n_stripes == 1 ? g->bins_in : g->stripe_bins[sIdx];
+      int64_t off = 0, goff = 0;
+      for (int b = 0; b < 4; ++b) {
+        A.n[b] = bins.count[b];
+        A.off[b] = off;
+        off += A.n[b];
+        A.goff[b] = goff;
+        A.grid[b] = bins.grid[b];
+        goff += A.grid[b];
+      }
+      A.bin_rows = bins.rows;
+      if (n_stripes == 1) {
+        A.sp_lo = g->in_row_ptr;
+        A.sp_hi = g->in_row_ptr + 1;
+        A.mode = 0;
+      } else {
+        A.sp_lo = g->stripe_ptr + (size_t)sIdx * rows;
+        A.sp_hi = g->stripe_ptr + (size_t)(sIdx + 1) * rows;
+        A.mode = sIdx == 0 ? 1 : (sIdx == n_stripes - 1 ? 3 : 2);
+      }
+      if (goff > 0) {
+        hipLaunchKernelGGL(k_pr_sweep, dim3((uint32_t)goff), dim3(kBlock), 0, ctx->stream,
+                           A);
+      }
+    }
     MGX_HIP_TRY(hipEventRecord(run->ev_stop[run->ev_used], ctx->stream));
     ++run->ev_used;
     MGX_HIP_TRY(hipGetLastError());
@@ -318,6 +351,10 @@ mgx_status pagerank_start_common(mgx_context *ctx, mgx_graph *g, double damping,
   }
   MGX_HIP_TRY(hipMalloc(&run->d_delta, sizeof(uint32_t)));
   MGX_HIP_TRY(hipMalloc(&run->d_scratch, (V + 1) * sizeof(double)));
+  if (g->n_stripes > 1) {
+    const int64_t rows = (dist ? (row_end < V ? row_end : V) : V) - row_begin;
+    MGX_HIP_TRY(hipMalloc(&run->d_partial, (rows > 0 ? rows : 1) * sizeof(double)));
+  }
   if (V > 0) {
     const float r0 = (float)(1.0 / (double)V);
     hipLaunchKernelGGL(k_pr_init, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
@@ -335,6 +372,7 @@ void pagerank_run_free(mgx_pagerank_run *run) {
   }
   if (run->d_delta) (void)hipFree(run->d_delta);
   if (run->d_scratch) (void)hipFree(run->d_scratch);
+  if (run->d_partial) (void)hipFree(run->d_partial);
   for (auto e : run->ev_start) (void)hipEventDestroy(e);
   for (auto e : run->ev_stop) (void)hipEventDestroy(e);
   delete run;
