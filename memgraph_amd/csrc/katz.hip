@@ -55,7 +55,28 @@ __device__ inline void katz_rows(const KatzArgs &A, int sec, int64_t block_in_se
     if (ri < nrows) {
       row = rows_list[ri];
       const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
-      for (uint32_t j = s + sub; j < e; j += LANES) acc += A.omega_old[A.col[j]];
+      if constexpr (LANES >= 64) {
+        // Same access shape as the PageRank sweep: scalar head to 16-B
+        // alignment, int4 nontemporal col loads, 4 gathers in flight.
+        uint32_t s_al = (s + 3u) & ~3u;
+        if (s_al > e) s_al = e;
+        for (uint32_t j = s + sub; j < s_al; j += LANES) acc += A.omega_old[A.col[j]];
+        const uint32_t nvec = (e - s_al) / 4;
+        typedef int v4i __attribute__((ext_vector_type(4)));
+        const v4i *col4 = reinterpret_cast<const v4i *>(A.col + s_al);
+        for (uint32_t c = sub; c < nvec; c += LANES) {
+          const v4i cc = __builtin_nontemporal_load(col4 + c);
+          acc += A.omega_old[cc.x];
+          acc += A.omega_old[cc.y];
+          acc += A.omega_old[cc.z];
+          acc += A.omega_old[cc.w];
+        }
+        for (uint32_t j = s_al + nvec * 4 + sub; j < e; j += LANES)
+          acc += A.omega_old[A.col[j]];
+      } else {
+        for (uint32_t j = s + sub; j < e; j += LANES)
+          acc += A.omega_old[__builtin_nontemporal_load(A.col + j)];
+      }
     }
     if constexpr (LANES <= 64) {
       for (int o = LANES / 2; o; o >>= 1) acc += __shfl_down(acc, o, LANES);

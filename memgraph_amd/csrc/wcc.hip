@@ -57,9 +57,25 @@ __device__ inline bool wcc_rows(const WccArgs &A, int sec, int64_t block_in_sec)
     if (ri < nrows) {
       row = rows_list[ri];
       const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
-      for (uint32_t j = s + sub; j < e; j += LANES) {
-        const int32_t l = A.label[A.col[j]];
-        if (l < m) m = l;
+      if constexpr (LANES >= 64) {
+        uint32_t s_al = (s + 3u) & ~3u;
+        if (s_al > e) s_al = e;
+        for (uint32_t j = s + sub; j < s_al; j += LANES) m = min(m, A.label[A.col[j]]);
+        const uint32_t nvec = (e - s_al) / 4;
+        typedef int v4i __attribute__((ext_vector_type(4)));
+        const v4i *col4 = reinterpret_cast<const v4i *>(A.col + s_al);
+        for (uint32_t c = sub; c < nvec; c += LANES) {
+          const v4i cc = __builtin_nontemporal_load(col4 + c);
+          m = min(m, A.label[cc.x]);
+          m = min(m, A.label[cc.y]);
+          m = min(m, A.label[cc.z]);
+          m = min(m, A.label[cc.w]);
+        }
+        for (uint32_t j = s_al + nvec * 4 + sub; j < e; j += LANES)
+          m = min(m, A.label[A.col[j]]);
+      } else {
+        for (uint32_t j = s + sub; j < e; j += LANES)
+          m = min(m, A.label[__builtin_nontemporal_load(A.col + j)]);
       }
     }
     if constexpr (LANES <= 64) {
