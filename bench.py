@@ -51,7 +51,74 @@ def parse_args():
     p.add_argument("--cpu-sample-iters", type=int, default=6)
     p.add_argument("--check", type=int, default=0,
                    help="verify the first steps against the oracle (small scales)")
+    p.add_argument("--algo", default="pagerank",
+                   choices=["pagerank", "wcc", "katz", "louvain"],
+                   help="pagerank is the contract workload; the others are "
+                        "secondary evidence lines (BASELINE.md configs 3-4)")
     return p.parse_args()
+
+
+def run_secondary(args):
+    """Secondary algorithms: whole-call runtime on
+    an RMAT graph, N=1 only. Emits one JSON line; not the contract metric."""
+    import numpy as np  # noqa: F401
+    from memgraph_amd.native import (BUILD_IN_CSR, BUILD_SYM_CSR, BUILD_WEIGHTED,
+                                     Native)
+    nat = Native()
+    if nat.device_count() == 0:
+        print(json.dumps({"error": "no HIP device"}))
+        sys.exit(1)
+    ctx = nat.init(0)
+    scale = args.scale if args.scale != 26 else 24  # config 3/4 default
+    V = 1 << scale
+    E = args.edge_factor * V
+    flags = {"wcc": BUILD_SYM_CSR, "katz": BUILD_IN_CSR,
+             "louvain": BUILD_SYM_CSR | BUILD_WEIGHTED}[args.algo]
+    g = nat.graph_rmat(ctx, scale, E, seed=args.seed, flags=flags)
+    reps = max(args.steps // 10, 1)
+    extra = {}
+    # one untimed warm call, then timed repetitions of the whole call
+    for phase in ("warm", "timed"):
+        if phase == "timed":
+            nat.sync(ctx)
+            t0 = time.perf_counter()
+        for _ in range(1 if phase == "warm" else reps):
+            if args.algo == "wcc":
+                _, n = nat.wcc(ctx, g, V)
+                extra["components"] = int(n)
+            elif args.algo == "katz":
+                _, iters = nat.katz(ctx, g, V)
+                extra["iterations"] = int(iters)
+            else:
+                _, n = nat.louvain(ctx, g, V)
+                extra["communities"] = int(n)
+        if phase == "timed":
+            nat.sync(ctx)
+            t1 = time.perf_counter()
+    secs = (t1 - t0) / reps
+    out = {
+        "metric": f"{args.algo} runtime",
+        "value": secs * 1e3,
+        "unit": "ms",
+        "n_gpus": 1,
+        "steps": reps,
+        "warmup": 1,
+        "ms_per_step": secs * 1e3,
+        "higher_is_better": False,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "f64" if args.algo == "katz" else "int32/f64",
+        "data": "synthetic",
+        "config": {"workload": f"RMAT-{scale} {args.algo}", "scale": scale,
+                   "vertices": V, "edges": E, "seed": args.seed,
+                   "parallelism": "single"},
+        "edges_per_s": E / secs,
+        "csr_build_ms": nat.graph_build_ms(g),
+        **extra,
+    }
+    print(json.dumps(out))
+    nat.graph_destroy(ctx, g)
+    nat.destroy(ctx)
 
 
 def cpu_baseline_leg(args):
@@ -100,6 +167,9 @@ def read_traffic(workload):
 
 def main():
     args = parse_args()
+    if args.algo != "pagerank":
+        run_secondary(args)
+        return
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
