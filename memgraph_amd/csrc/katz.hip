@@ -112,6 +112,13 @@ __global__ void __launch_bounds__(kBlock) k_katz_sweep(KatzArgs A) {
   }
 }
 
+__global__ void k_scatter_f64(int64_t n, const double *in, const int32_t *order,
+                              double *out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[order ? order[i] : i] = in[i];
+}
+
 __global__ void k_fill_f64(int64_t n, double v, double *p) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
@@ -122,6 +129,20 @@ __global__ void k_iota_u32(int64_t n, uint32_t *p) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
     p[i] = (uint32_t)i;
+}
+
+// Stage the convergence-sort input in ORIGINAL vertex order so the stable
+// descending radix sort breaks centrality ties by original id — the
+// oracle's (documented) tie rule — while values stay permuted indices for
+// the lr/ur lookups.
+__global__ void k_sort_stage(int64_t n, const double *cent, const int32_t *order,
+                             double *keys, uint32_t *vals) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t old = order ? order[i] : i;
+    keys[old] = cent[i];
+    vals[old] = (uint32_t)i;
+  }
 }
 
 __global__ void k_max_u32(int64_t n, const uint32_t *x, uint32_t *out) {
@@ -229,12 +250,11 @@ mgx_status mgx_katz_impl(mgx_context *ctx, mgx_graph *g, double alpha, double ep
     hipLaunchKernelGGL(k_katz_sweep, dim3((uint32_t)goff), dim3(kBlock), 0, ctx->stream, A);
     cur = 1 - cur;
 
-    // Stable descending sort of (centrality, id): radix_sort_pairs_desc is
-    // stable, and vals_in is ascending, so ties order by node id.
-    MGX_HIP_TRY(hipMemcpyAsync(keys_in, cent, V * sizeof(double),
-                               hipMemcpyDeviceToDevice, ctx->stream));
-    hipLaunchKernelGGL(k_iota_u32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0,
-                       ctx->stream, V, vals_in);
+    // Stable descending sort of (centrality, original id): the stage kernel
+    // places entries in original-id order, so stability gives the oracle's
+    // tie rule; values are permuted indices for the lr/ur lookups.
+    hipLaunchKernelGGL(k_sort_stage, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0,
+                       ctx->stream, V, cent, g->order, keys_in, vals_in);
     size_t tmp_bytes = 0;
     auto err = rocprim::radix_sort_pairs_desc(nullptr, tmp_bytes, keys_in, keys_out,
                                               vals_in, vals_out, V, 0, 64, ctx->stream);
@@ -261,7 +281,12 @@ mgx_status mgx_katz_impl(mgx_context *ctx, mgx_graph *g, double alpha, double ep
   }
 
   if (status == MGX_OK && out_centrality) {
-    MGX_HIP_TRY(hipMemcpyAsync(out_centrality, cent, V * sizeof(double),
+    // in-CSR lives in the hot-first permuted space: scatter back to
+    // original ids on device, then one contiguous D2H.
+    double *scat = keys_in;  // reuse
+    hipLaunchKernelGGL(k_scatter_f64, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0,
+                       ctx->stream, V, cent, g->order, scat);
+    MGX_HIP_TRY(hipMemcpyAsync(out_centrality, scat, V * sizeof(double),
                                hipMemcpyDeviceToHost, ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   }

@@ -249,7 +249,8 @@ extern "C" mgx_status mgx_graph_destroy(mgx_context *ctx, mgx_graph *g) {
   if (g->bins_in.rows) (void)hipFree(g->bins_in.rows);
   if (g->bins_sym.rows) (void)hipFree(g->bins_sym.rows);
   if (g->stripe_ptr) (void)hipFree(g->stripe_ptr);
-  for (int i = 0; i < 8; ++i) {
+  if (g->order) (void)hipFree(g->order);
+  for (int i = 0; i < 16; ++i) {
     if (g->stripe_bins[i].rows) (void)hipFree(g->stripe_bins[i].rows);
   }
   delete g;

@@ -71,9 +71,16 @@ struct mgx_graph {
   int64_t in_edges = 0;  // == n_edges unless sharded
   int64_t row_begin = 0, row_end = 0;  // == [0, n_vertices) unless sharded
 
-  // out-degree data (always global, all vertices).
+  // out-degree data (always global, all vertices). When `order` is set,
+  // the in-CSR lives in a HOT-FIRST PERMUTED vertex space: vertices are
+  // renumbered by descending out-degree so the most-gathered contrib
+  // entries pack into the lowest addresses (L2/L3-resident under power-law
+  // skew). order[new] = original id; out_degree/inv_outdeg are in the
+  // permuted space. Outputs are scattered back to original ids at download.
+  // Sharded graphs keep the identity layout (contiguous dst ranges).
   uint32_t *out_degree = nullptr;
   float *inv_outdeg = nullptr;
+  int32_t *order = nullptr;  // [V] new -> original, or nullptr (identity)
 
   // symmetric CSR (WCC/Louvain): each input edge twice.
   uint32_t *sym_row_ptr = nullptr;
@@ -91,7 +98,7 @@ struct mgx_graph {
   int n_stripes = 1;
   int64_t stripe_width = 0;
   uint32_t *stripe_ptr = nullptr;          // [(n_stripes+1) * rows]
-  mgx_bins stripe_bins[8];                 // per stripe (max 8 stripes)
+  mgx_bins stripe_bins[16];                // per stripe (max 16)
 
   double build_ms = 0.0;
 };

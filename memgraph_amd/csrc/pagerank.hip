@@ -174,11 +174,14 @@ __global__ void k_sum_f32(int64_t n, const float *x, double *out) {
   }
 }
 
-__global__ void k_pr_widen(int64_t n, const float *rank, const double *sum, double *out) {
+__global__ void k_pr_widen(int64_t n, const float *rank, const double *sum,
+                           const int32_t *order, double *out) {
+  // order: permuted -> original vertex id (hot-first layout); identity when
+  // null. Scatter on device so the D2H stays one contiguous copy.
   const double inv = 1.0 / *sum;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
-    out[i] = (double)rank[i] * inv;
+    out[order ? order[i] : i] = (double)rank[i] * inv;
 }
 
 inline int64_t grid_for(int64_t work, int64_t cap = 4096) {
@@ -316,7 +319,8 @@ mgx_status mgx_pagerank_normalize_download(mgx_pagerank_run *run, double *out_ra
   hipLaunchKernelGGL(k_sum_f32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
                      V, run->rank[run->cur], run->d_scratch);
   hipLaunchKernelGGL(k_pr_widen, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
-                     V, run->rank[run->cur], run->d_scratch, run->d_scratch + 1);
+                     V, run->rank[run->cur], run->d_scratch, run->g->order,
+                     run->d_scratch + 1);
   if (out_rank) {
     MGX_HIP_TRY(hipMemcpyAsync(out_rank, run->d_scratch + 1, V * sizeof(double),
                                hipMemcpyDeviceToHost, ctx->stream));
