@@ -89,6 +89,8 @@ enum mgp_error mgp_value_get_string(struct mgp_value *val, const char **result);
 enum mgp_error mgp_value_get_list(struct mgp_value *val, struct mgp_list **result); /* :451 */
 enum mgp_error mgp_value_get_vertex(struct mgp_value *val,
                                     struct mgp_vertex **result);            /* :461 */
+enum mgp_error mgp_value_get_edge(struct mgp_value *val,
+                                  struct mgp_edge **result);                /* :466 */
 
 /* --- lists (mg_procedure.h:554-563) --- */
 enum mgp_error mgp_list_size(struct mgp_list *list, size_t *result);        /* :554 */

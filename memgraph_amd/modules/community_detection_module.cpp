@@ -9,11 +9,13 @@
 //                           coloring_alg_threshold=0.01:float,
 //                           num_of_threads=<cores/2>:int)
 //   -> (node: node, community_id: int)
+// plus community_detection.get_subgraph(subgraph_nodes: list<node>,
+// subgraph_relationships: list<relationship>, <same optional args>)
+// (community_detection_module.cpp:136-152 / OnSubgraph).
 // coloring / min_graph_shrink / coloring_alg_threshold / num_of_threads are
 // accepted for drop-in compatibility; the GPU path always runs the basic
 // (non-coloring) algorithm, minGraphSize is unused by the reference's basic
 // path too (runMultiPhaseBasic ignores it), and thread count is N/A on GPU.
-// The .get_subgraph variant is a §8f "next" row (not in round 1).
 
 #include <thread>
 
@@ -29,6 +31,32 @@ constexpr const char *kFieldCommunity = "community_id";
 constexpr const char *kDefaultWeightProperty = "weight";
 constexpr double kDefaultWeight = 1.0;
 
+void RunLouvain(mgp_graph *graph, mgp_result *result, mgp_memory *memory,
+                const ScanResult &scan, double threshold) {
+  const int64_t V = (int64_t)scan.dense_to_mg.size();
+  // The reference emits nothing when the scanned graph has no edges
+  // (community_detection_module.cpp:72-74).
+  if (V == 0 || scan.src.empty()) return;
+
+  mgx_context *ctx = Ctx();
+  GraphGuard gg{ctx};
+  CheckMgx(mgx_graph_from_coo(ctx, scan.src.data(), scan.dst.data(),
+                              scan.weights.data(), V, (int64_t)scan.src.size(),
+                              MGX_BUILD_SYM_CSR | MGX_BUILD_WEIGHTED, &gg.g),
+           "mgx_graph_from_coo");
+  std::vector<int64_t> community(V);
+  int64_t n_communities = 0;
+  CheckMgx(mgx_louvain(ctx, gg.g, threshold, community.data(), &n_communities),
+           "mgx_louvain");
+
+  for (int64_t v = 0; v < V; ++v) {
+    EmitNodeRecord(graph, result, memory, scan.dense_to_mg[v], kFieldNode,
+                   [&](mgp_result_record *rec) {
+                     InsertInt(rec, kFieldCommunity, community[v], memory);
+                   });
+  }
+}
+
 void OnGraph(mgp_list *args, mgp_graph *graph, mgp_result *result, mgp_memory *memory) {
   try {
     const char *weight_property = ArgString(args, 0);
@@ -39,28 +67,34 @@ void OnGraph(mgp_list *args, mgp_graph *graph, mgp_result *result, mgp_memory *m
     // observable community numbering depends on it.
     ScanResult scan = ScanGraph(graph, memory, Numbering::kFirstSeen,
                                 /*read_weights=*/true, weight_property, kDefaultWeight);
-    const int64_t V = (int64_t)scan.dense_to_mg.size();
-    // The reference emits nothing when the scanned graph has no edges
-    // (community_detection_module.cpp:72-74).
-    if (V == 0 || scan.src.empty()) return;
+    RunLouvain(graph, result, memory, scan, threshold);
+  } catch (const std::exception &e) {
+    (void)mgp_result_set_error_msg(result, e.what());
+    return;
+  }
+}
 
-    mgx_context *ctx = Ctx();
-    GraphGuard gg{ctx};
-    CheckMgx(mgx_graph_from_coo(ctx, scan.src.data(), scan.dst.data(),
-                                scan.weights.data(), V, (int64_t)scan.src.size(),
-                                MGX_BUILD_SYM_CSR | MGX_BUILD_WEIGHTED, &gg.g),
-             "mgx_graph_from_coo");
-    std::vector<int64_t> community(V);
-    int64_t n_communities = 0;
-    CheckMgx(mgx_louvain(ctx, gg.g, threshold, community.data(), &n_communities),
-             "mgx_louvain");
+mgp_list *ArgList(mgp_list *args, size_t i) {
+  mgp_value *v = nullptr;
+  Check(mgp_list_at(args, i, &v), "list_at");
+  mgp_list *out = nullptr;
+  Check(mgp_value_get_list(v, &out), "value_get_list");
+  return out;
+}
 
-    for (int64_t v = 0; v < V; ++v) {
-      EmitNodeRecord(graph, result, memory, scan.dense_to_mg[v], kFieldNode,
-                     [&](mgp_result_record *rec) {
-                       InsertInt(rec, kFieldCommunity, community[v], memory);
-                     });
-    }
+void OnSubgraph(mgp_list *args, mgp_graph *graph, mgp_result *result, mgp_memory *memory) {
+  try {
+    // Argument layout of the reference's get_subgraph
+    // (LouvainCommunityDetection with subgraph=true: the two lists first,
+    // then the same optionals — community_detection_module.cpp:51-64).
+    mgp_list *nodes = ArgList(args, 0);
+    mgp_list *relationships = ArgList(args, 1);
+    const char *weight_property = ArgString(args, 2);
+    const double threshold = ArgDouble(args, 5);
+    ScanResult scan = ScanSubgraph(graph, memory, nodes, relationships,
+                                   /*read_weights=*/true, weight_property,
+                                   kDefaultWeight);
+    RunLouvain(graph, result, memory, scan, threshold);
   } catch (const std::exception &e) {
     (void)mgp_result_set_error_msg(result, e.what());
     return;
@@ -104,6 +138,27 @@ extern "C" int mgp_init_module(struct mgp_module *module, struct mgp_memory *mem
 
     Check(mgp_proc_add_result(proc, kFieldNode, t_node), "add_result");
     Check(mgp_proc_add_result(proc, kFieldCommunity, t_int), "add_result");
+
+    // get_subgraph (community_detection_module.cpp:136-152)
+    mgp_proc *sproc = nullptr;
+    Check(mgp_module_add_read_procedure(module, "get_subgraph", OnSubgraph, &sproc),
+          "add_read_procedure(get_subgraph)");
+    mgp_type *t_rel = nullptr, *t_list_node = nullptr, *t_list_rel = nullptr;
+    Check(mgp_type_relationship(&t_rel), "type_relationship");
+    Check(mgp_type_list(t_node, &t_list_node), "type_list(node)");
+    Check(mgp_type_list(t_rel, &t_list_rel), "type_list(rel)");
+    Check(mgp_proc_add_arg(sproc, "subgraph_nodes", t_list_node), "arg");
+    Check(mgp_proc_add_arg(sproc, "subgraph_relationships", t_list_rel), "arg");
+    Check(mgp_proc_add_opt_arg(sproc, "weight_property", t_string, d_weight_prop), "arg");
+    Check(mgp_proc_add_opt_arg(sproc, "coloring", t_bool, d_coloring), "arg");
+    Check(mgp_proc_add_opt_arg(sproc, "min_graph_shrink", t_int, d_shrink), "arg");
+    Check(mgp_proc_add_opt_arg(sproc, "community_alg_threshold", t_float, d_threshold),
+          "arg");
+    Check(mgp_proc_add_opt_arg(sproc, "coloring_alg_threshold", t_float, d_col_threshold),
+          "arg");
+    Check(mgp_proc_add_opt_arg(sproc, "num_of_threads", t_int, d_threads), "arg");
+    Check(mgp_proc_add_result(sproc, kFieldNode, t_node), "add_result");
+    Check(mgp_proc_add_result(sproc, kFieldCommunity, t_int), "add_result");
   } catch (const std::exception &) {
     if (d_weight_prop) mgp_value_destroy(d_weight_prop);
     if (d_coloring) mgp_value_destroy(d_coloring);

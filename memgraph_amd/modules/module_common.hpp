@@ -167,6 +167,72 @@ inline ScanResult ScanGraph(mgp_graph *graph, mgp_memory *memory, Numbering numb
   return out;
 }
 
+// Subgraph scan mirroring louvain_alg::GetLouvainSubgraph
+// (louvain.cpp:120-156): nodes list interned first (first-seen dense ids,
+// duplicates deduped), then edges appended PER LIST ENTRY (duplicates
+// create multi-edges, exactly as the reference does) when both endpoints
+// are in the node set.
+inline ScanResult ScanSubgraph(mgp_graph *graph, mgp_memory *memory, mgp_list *nodes,
+                               mgp_list *edges, bool read_weights,
+                               const char *weight_property, double default_weight) {
+  ScanResult out;
+  std::unordered_map<int64_t, int64_t> mg_to_dense;
+  size_t n_nodes = 0, n_edges = 0;
+  Check(mgp_list_size(nodes, &n_nodes), "list_size(nodes)");
+  Check(mgp_list_size(edges, &n_edges), "list_size(edges)");
+  for (size_t i = 0; i < n_nodes; ++i) {
+    if (mgp_must_abort(graph)) throw MgpError("query aborted");
+    mgp_value *v = nullptr;
+    Check(mgp_list_at(nodes, i, &v), "list_at(nodes)");
+    mgp_vertex *vert = nullptr;
+    Check(mgp_value_get_vertex(v, &vert), "value_get_vertex");
+    mgp_vertex_id vid{0};
+    Check(mgp_vertex_get_id(vert, &vid), "vertex_get_id");
+    if (!mg_to_dense.count(vid.as_int)) {
+      mg_to_dense.emplace(vid.as_int, (int64_t)out.dense_to_mg.size());
+      out.dense_to_mg.push_back(vid.as_int);
+    }
+  }
+  for (size_t i = 0; i < n_edges; ++i) {
+    if (mgp_must_abort(graph)) throw MgpError("query aborted");
+    mgp_value *v = nullptr;
+    Check(mgp_list_at(edges, i, &v), "list_at(edges)");
+    mgp_edge *e = nullptr;
+    Check(mgp_value_get_edge(v, &e), "value_get_edge");
+    mgp_vertex *from = nullptr, *to = nullptr;
+    Check(mgp_edge_get_from(e, &from), "edge_get_from");
+    Check(mgp_edge_get_to(e, &to), "edge_get_to");
+    mgp_vertex_id fid{0}, tid{0};
+    Check(mgp_vertex_get_id(from, &fid), "vertex_get_id(from)");
+    Check(mgp_vertex_get_id(to, &tid), "vertex_get_id(to)");
+    auto fit = mg_to_dense.find(fid.as_int);
+    auto tit = mg_to_dense.find(tid.as_int);
+    if (fit == mg_to_dense.end() || tit == mg_to_dense.end()) continue;
+    out.src.push_back(fit->second);
+    out.dst.push_back(tit->second);
+    if (read_weights) {
+      double w = default_weight;
+      mgp_value *pv = nullptr;
+      if (mgp_edge_get_property(e, weight_property, memory, &pv) == MGP_ERROR_NO_ERROR &&
+          pv) {
+        int is = 0;
+        int64_t iv = 0;
+        double dv = 0.0;
+        if (mgp_value_is_double(pv, &is) == MGP_ERROR_NO_ERROR && is &&
+            mgp_value_get_double(pv, &dv) == MGP_ERROR_NO_ERROR) {
+          w = dv;
+        } else if (mgp_value_is_int(pv, &is) == MGP_ERROR_NO_ERROR && is &&
+                   mgp_value_get_int(pv, &iv) == MGP_ERROR_NO_ERROR) {
+          w = (double)iv;
+        }
+        mgp_value_destroy(pv);
+      }
+      out.weights.push_back(w);
+    }
+  }
+  return out;
+}
+
 // Emission mirroring InsertPagerankRecord / InsertWeaklyComponentResult
 // ownership (mg_utils.hpp:256-316): nullptr vertex in non-transactional
 // storage is skipped, in transactional storage it is an error.

@@ -71,6 +71,21 @@ class ModuleHost:
             else:
                 self.mock.mock_add_edge(a, b)
 
+    def override_arg_node_list(self, pos, mg_ids):
+        import numpy as np
+        a = np.ascontiguousarray(mg_ids, dtype=np.int64)
+        self.mock.mock_override_arg_node_list(
+            ctypes.c_int64(pos), a.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            ctypes.c_int64(len(a)))
+
+    def override_arg_edge_list(self, pos, from_ids, to_ids):
+        import numpy as np
+        f = np.ascontiguousarray(from_ids, dtype=np.int64)
+        t = np.ascontiguousarray(to_ids, dtype=np.int64)
+        self.mock.mock_override_arg_edge_list(
+            ctypes.c_int64(pos), f.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            t.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)), ctypes.c_int64(len(f)))
+
     def override_arg(self, pos, value):
         if isinstance(value, bool):
             raise NotImplementedError

@@ -76,12 +76,13 @@ struct mgp_edge {
 
 namespace {
 struct Value {
-  enum Kind { kNull, kBool, kInt, kDouble, kString, kVertex, kList } kind = kNull;
+  enum Kind { kNull, kBool, kInt, kDouble, kString, kVertex, kEdge, kList } kind = kNull;
   int64_t i = 0;
   double d = 0.0;
   std::string s;
   mgp_vertex *vertex = nullptr;  // owned when kind == kVertex
-  struct mgp_list *list = nullptr;
+  struct mgp_edge *edge = nullptr;   // arena-owned
+  struct mgp_list *list = nullptr;   // arena-owned
 };
 }  // namespace
 
@@ -147,6 +148,19 @@ struct mgp_module {
 namespace {
 mgp_module g_module;
 mgp_memory g_memory;
+// Arena for list/edge/value objects handed to procedures via arg overrides;
+// cleared on mock_reset*/mock_call completion boundaries.
+std::vector<std::unique_ptr<mgp_list>> g_arena_lists;
+std::vector<std::unique_ptr<mgp_value>> g_arena_values;
+std::vector<std::unique_ptr<mgp_edge>> g_arena_edges;
+std::vector<std::unique_ptr<mgp_vertex>> g_arena_vertices;
+
+void clear_arena() {
+  g_arena_lists.clear();
+  g_arena_values.clear();
+  g_arena_edges.clear();
+  g_arena_vertices.clear();
+}
 mgp_graph g_graph_handle{&g_graph};
 mgp_result g_result_handle{&g_result};
 std::vector<std::pair<size_t, Value>> g_arg_overrides;
@@ -255,6 +269,12 @@ enum mgp_error mgp_value_get_list(struct mgp_value *val, struct mgp_list **resul
 enum mgp_error mgp_value_get_vertex(struct mgp_value *val, struct mgp_vertex **result) {
   if (val->v.kind != Value::kVertex) return MGP_ERROR_LOGIC_ERROR;
   *result = val->v.vertex;
+  return MGP_ERROR_NO_ERROR;
+}
+
+enum mgp_error mgp_value_get_edge(struct mgp_value *val, struct mgp_edge **result) {
+  if (val->v.kind != Value::kEdge) return MGP_ERROR_LOGIC_ERROR;
+  *result = val->v.edge;
   return MGP_ERROR_NO_ERROR;
 }
 
@@ -472,12 +492,14 @@ void mock_reset(void) {
   g_result = MockResult{};
   g_module.procs.clear();
   g_arg_overrides.clear();
+  clear_arena();
 }
 
 void mock_reset_graph(void) {
   g_graph = MockGraph{};
   g_result = MockResult{};
   g_arg_overrides.clear();
+  clear_arena();
 }
 
 void mock_add_vertex(int64_t mg_id) {
@@ -560,6 +582,59 @@ void mock_override_arg_double(int64_t pos, double v) {
   val.kind = Value::kDouble;
   val.d = v;
   g_arg_overrides.emplace_back((size_t)pos, val);
+}
+
+// Override an argument with a list of node values (by memgraph id).
+void mock_override_arg_node_list(int64_t pos, const int64_t *ids, int64_t n) {
+  auto list = std::make_unique<mgp_list>();
+  for (int64_t i = 0; i < n; ++i) {
+    auto vert = std::make_unique<mgp_vertex>();
+    vert->mg_id = ids[i];
+    auto val = std::make_unique<mgp_value>();
+    val->v.kind = Value::kVertex;
+    val->v.vertex = vert.get();
+    list->items.push_back(val.get());
+    g_arena_vertices.push_back(std::move(vert));
+    g_arena_values.push_back(std::move(val));
+  }
+  Value lv;
+  lv.kind = Value::kList;
+  lv.list = list.get();
+  g_arena_lists.push_back(std::move(list));
+  g_arg_overrides.emplace_back((size_t)pos, lv);
+}
+
+// Override an argument with a list of relationship values given as
+// (from, to) memgraph-id pairs; properties come from the first matching
+// graph edge (if any).
+void mock_override_arg_edge_list(int64_t pos, const int64_t *from_ids,
+                                 const int64_t *to_ids, int64_t n) {
+  auto list = std::make_unique<mgp_list>();
+  for (int64_t i = 0; i < n; ++i) {
+    auto edge = std::make_unique<mgp_edge>();
+    static MockEdge fallback;
+    const MockEdge *src_edge = &fallback;
+    for (const auto &e : g_graph.edges) {
+      if (e.from_mg == from_ids[i] && e.to_mg == to_ids[i]) {
+        src_edge = &e;
+        break;
+      }
+    }
+    edge->edge = src_edge;
+    edge->from_v.mg_id = from_ids[i];
+    edge->to_v.mg_id = to_ids[i];
+    auto val = std::make_unique<mgp_value>();
+    val->v.kind = Value::kEdge;
+    val->v.edge = edge.get();
+    list->items.push_back(val.get());
+    g_arena_edges.push_back(std::move(edge));
+    g_arena_values.push_back(std::move(val));
+  }
+  Value lv;
+  lv.kind = Value::kList;
+  lv.list = list.get();
+  g_arena_lists.push_back(std::move(list));
+  g_arg_overrides.emplace_back((size_t)pos, lv);
 }
 
 void mock_override_arg_string(int64_t pos, const char *v) {

@@ -83,6 +83,26 @@ def test_community_module_no_edges_emits_nothing():
     assert rows == []
 
 
+def test_community_get_subgraph():
+    # Mirrors tests/mage/e2e/community_detection_test/test_subgraph: two
+    # triangles + bridge node 6; MATCH (a)-[e]-(b) with id<6 collects each
+    # node per incident edge and each edge twice (both directions) — the
+    # reference dedups nodes but keeps duplicate edges as multi-edges.
+    h = ModuleHost("community_detection")
+    edges = [(0, 1), (1, 2), (2, 0), (3, 4), (4, 5), (5, 3), (5, 6)]
+    h.load_graph(list(range(7)), [e[0] for e in edges], [e[1] for e in edges])
+    sub_edges = [e for e in edges if e[0] < 6 and e[1] < 6]
+    nodes, elist = [], []
+    for a, b in sub_edges:
+        nodes += [a, b]
+        elist += [(a, b), (a, b)]  # undirected match yields the edge twice
+    h.override_arg_node_list(0, nodes)
+    h.override_arg_edge_list(1, [e[0] for e in elist], [e[1] for e in elist])
+    rows = h.call("get_subgraph")
+    got = {h.row_int(i, "node"): h.row_int(i, "community_id") for i in rows}
+    assert got == {0: 0, 1: 0, 2: 0, 3: 1, 4: 1, 5: 1}
+
+
 def test_pagerank_module_uniform_10k(oracle):
     # the full drop-in path (scan -> GPU -> emission)
     # on the 10k/50k uniform graph, vs the oracle.

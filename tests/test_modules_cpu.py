@@ -76,6 +76,13 @@ def test_community_registration():
         ["min_graph_shrink", "int"], ["community_alg_threshold", "float"],
         ["coloring_alg_threshold", "float"], ["num_of_threads", "int"]]
     assert procs["get"]["results"] == [["node", "node"], ["community_id", "int"]]
+    # get_subgraph (community_detection_module.cpp:136-152)
+    assert procs["get_subgraph"]["args"] == [
+        ["subgraph_nodes", "list"], ["subgraph_relationships", "list"],
+        ["weight_property", "string"], ["coloring", "bool"],
+        ["min_graph_shrink", "int"], ["community_alg_threshold", "float"],
+        ["coloring_alg_threshold", "float"], ["num_of_threads", "int"]]
+    assert procs["get_subgraph"]["results"] == [["node", "node"], ["community_id", "int"]]
 
 
 FAIL_LOUD_SCRIPT = r"""
