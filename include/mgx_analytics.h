@@ -56,6 +56,7 @@ mgx_status mgx_sync(mgx_context *ctx);
 #define MGX_BUILD_IN_CSR 1u
 #define MGX_BUILD_SYM_CSR 2u
 #define MGX_BUILD_WEIGHTED 4u /* with SYM_CSR: carry fp32 edge weights */
+#define MGX_BUILD_OUT_CSR 8u /* row = source, cols = destinations (Brandes) */
 
 /* Upload a host COO and build the requested CSRs on device.
  * Replaces the layouts built by pagerank_alg::PageRankGraph
@@ -145,6 +146,14 @@ mgx_status mgx_katz(mgx_context *ctx, mgx_graph *g, double alpha, double epsilon
  * community -1 exactly when the reference would. */
 mgx_status mgx_louvain(mgx_context *ctx, mgx_graph *g, double threshold,
                        int64_t *out_community, int64_t *n_communities);
+
+/* ---- Betweenness centrality (replaces
+ *      betweenness_centrality_alg::BetweennessCentrality,
+ *      betweenness_centrality_module/algorithm/betweenness_centrality.cpp
+ *      :71-147 — exact Brandes, O(V*E)) -----------------------------------
+ * directed needs MGX_BUILD_OUT_CSR; undirected needs MGX_BUILD_SYM_CSR. */
+mgx_status mgx_betweenness(mgx_context *ctx, mgx_graph *g, int directed, int normalize,
+                           double *out_bc);
 
 /* ---- multi-GPU (RCCL over xGMI) ------------------------- */
 

@@ -588,6 +588,21 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     MGX_TRY(mgx_build_stripes(ctx, g));
   }
 
+  if (flags & MGX_BUILD_OUT_CSR) {
+    // out-CSR in the ORIGINAL vertex space (Brandes keys results by scan
+    // ids and traverses forward edges; no hot-first perm here).
+    MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
+    hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
+                       counts);
+    MGX_HIP_TRY(hipMalloc(&g->out_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_TRY(scan_counts(ctx, counts, V, g->out_row_ptr));
+    MGX_HIP_TRY(hipMalloc(&g->out_col, (E > 0 ? E : 1) * sizeof(int32_t)));
+    // reuse the sorted-cols builder with (src,dst) swapped: rows = sources
+    MGX_TRY(build_sorted_cols(ctx, d_dst, d_src, nullptr, E, V, false, 0, 0, g->out_col,
+                              E));
+    MGX_TRY(mgx_build_bins(ctx, g->out_row_ptr, V, &g->bins_out));
+  }
+
   if (flags & MGX_BUILD_SYM_CSR) {
     MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
     hipLaunchKernelGGL(k_hist2, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,

@@ -243,6 +243,9 @@ extern "C" mgx_status mgx_graph_destroy(mgx_context *ctx, mgx_graph *g) {
   if (g->in_col) (void)hipFree(g->in_col);
   if (g->out_degree) (void)hipFree(g->out_degree);
   if (g->inv_outdeg) (void)hipFree(g->inv_outdeg);
+  if (g->out_row_ptr) (void)hipFree(g->out_row_ptr);
+  if (g->out_col) (void)hipFree(g->out_col);
+  if (g->bins_out.rows) (void)hipFree(g->bins_out.rows);
   if (g->sym_row_ptr) (void)hipFree(g->sym_row_ptr);
   if (g->sym_col) (void)hipFree(g->sym_col);
   if (g->sym_w) (void)hipFree(g->sym_w);
@@ -277,4 +280,10 @@ extern "C" mgx_status mgx_louvain(mgx_context *ctx, mgx_graph *g, double thresho
                                   int64_t *out_community, int64_t *n_communities) {
   MGX_HIP_TRY(hipSetDevice(ctx->device));
   return mgx_louvain_impl(ctx, g, threshold, out_community, n_communities);
+}
+
+extern "C" mgx_status mgx_betweenness(mgx_context *ctx, mgx_graph *g, int directed,
+                                      int normalize, double *out_bc) {
+  MGX_HIP_TRY(hipSetDevice(ctx->device));
+  return mgx_betweenness_impl(ctx, g, directed, normalize, out_bc);
 }

@@ -82,6 +82,11 @@ struct mgx_graph {
   float *inv_outdeg = nullptr;
   int32_t *order = nullptr;  // [V] new -> original, or nullptr (identity)
 
+  // out-CSR (row = source, cols = destinations; identity layout) for
+  // directed Brandes BFS.
+  uint32_t *out_row_ptr = nullptr;
+  int32_t *out_col = nullptr;
+
   // symmetric CSR (WCC/Louvain): each input edge twice.
   uint32_t *sym_row_ptr = nullptr;
   int32_t *sym_col = nullptr;
@@ -89,6 +94,7 @@ struct mgx_graph {
 
   mgx_bins bins_in;   // over in-CSR rows
   mgx_bins bins_sym;  // over sym-CSR rows
+  mgx_bins bins_out;  // over out-CSR rows
 
   // Source-striped in-CSR view (PageRank at large V): cols are sorted
   // within each row, so stripe s of row v is the contiguous sub-range
@@ -168,6 +174,9 @@ mgx_status mgx_katz_impl(mgx_context *ctx, mgx_graph *g, double alpha, double ep
 // louvain.hip
 mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
                             int64_t *out_community, int64_t *n_communities);
+// betweenness.hip
+mgx_status mgx_betweenness_impl(mgx_context *ctx, mgx_graph *g, int directed, int normalize,
+                                double *out_bc);
 
 // comm.cpp (RCCL)
 mgx_status mgx_comm_allgather_f32(mgx_context *ctx, const float *send, float *recv,

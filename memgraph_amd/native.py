@@ -14,6 +14,7 @@ LIB_PATH = os.path.join(_DIR, "lib", "libmgx_analytics.so")
 BUILD_IN_CSR = 1
 BUILD_SYM_CSR = 2
 BUILD_WEIGHTED = 4
+BUILD_OUT_CSR = 8
 
 UNIQUE_ID_BYTES = 128
 
@@ -188,6 +189,14 @@ class Native:
             self.lib.mgx_louvain(ctx, g, ctypes.c_double(threshold),
                                  out.ctypes.data_as(_I64), ctypes.byref(n)), "mgx_louvain")
         return out, n.value
+
+    def betweenness(self, ctx, g, n_vertices, directed=True, normalize=True):
+        out = np.zeros(n_vertices, dtype=np.float64)
+        self._check(
+            self.lib.mgx_betweenness(ctx, g, ctypes.c_int(1 if directed else 0),
+                                     ctypes.c_int(1 if normalize else 0),
+                                     out.ctypes.data_as(_F64)), "mgx_betweenness")
+        return out
 
     # --- comm ---
     def comm_unique_id(self):

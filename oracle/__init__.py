@@ -94,6 +94,17 @@ class Oracle(_Lib):
         return self._fn("oracle_modularity", ctypes.c_double)(
             ctypes.c_int64(n_vertices), ctypes.c_int64(len(src)), psrc, pdst, pw, pc)
 
+    def betweenness(self, n_vertices, src, dst, directed=True, normalize=True):
+        src, psrc = _as_i64(src)
+        dst, pdst = _as_i64(dst)
+        out = np.zeros(n_vertices, dtype=np.float64)
+        rc = self._fn("oracle_betweenness", ctypes.c_int64)(
+            ctypes.c_int64(n_vertices), ctypes.c_int64(len(src)), psrc, pdst,
+            ctypes.c_int32(1 if directed else 0), ctypes.c_int32(1 if normalize else 0),
+            out.ctypes.data_as(_F64))
+        assert rc == 0, "oracle_betweenness failed"
+        return out
+
     def gen_rmat(self, scale, n_edges, seed=1, a=0.57, b=0.19, c=0.19):
         src = np.zeros(n_edges, dtype=np.int64)
         dst = np.zeros(n_edges, dtype=np.int64)
@@ -156,6 +167,18 @@ class Reference(_Lib):
             ctypes.c_int64(iterations), ctypes.c_double(damping), ctypes.c_int64(n_threads),
             out.ctypes.data_as(_F64))
         return out, secs
+
+    def betweenness(self, n_vertices, src, dst, directed=True, normalize=True,
+                    n_threads=1):
+        src, psrc = _as_i64(src)
+        dst, pdst = _as_i64(dst)
+        out = np.zeros(n_vertices, dtype=np.float64)
+        rc = self._fn("ref_betweenness", ctypes.c_int64)(
+            ctypes.c_int64(n_vertices), ctypes.c_int64(len(src)), psrc, pdst,
+            ctypes.c_int32(1 if directed else 0), ctypes.c_int32(1 if normalize else 0),
+            ctypes.c_int64(n_threads), out.ctypes.data_as(_F64))
+        assert rc == 0
+        return out
 
     def katz(self, n_vertices, src, dst, alpha=0.2, epsilon=1e-2):
         src, psrc = _as_i64(src)

@@ -40,6 +40,7 @@ SUITES = {
     "weakly_connected_components_test": "wcc",
     "community_detection_test": "community_detection",
     "katz_test": "katz",
+    "betweenness_centrality_test": "betweenness",
 }
 SKIP_SUBSTR = ("cugraph", "online", "subgraph")
 
@@ -78,7 +79,11 @@ def parse_query(query):
     if raw_args:
         for tok in raw_args.split(","):
             tok = tok.strip()
-            if tok.startswith('"') or tok.startswith("'"):
+            if tok in ("True", "true", "TRUE"):
+                args.append(True)
+            elif tok in ("False", "false", "FALSE"):
+                args.append(False)
+            elif tok.startswith('"') or tok.startswith("'"):
                 args.append(tok.strip("\"'"))
             elif re.fullmatch(r"-?\d+", tok):
                 args.append(int(tok))
@@ -117,6 +122,21 @@ def check_community(o, ref, fx):
         if pid in exp:
             assert c_o[dense] == exp[pid], (fx["name"], pid, c_o[dense], exp[pid])
     _ = weight_prop_given
+
+
+def check_betweenness(o, ref, fx):
+    args = fx["args"]
+    directed = bool(args[0]) if len(args) > 0 else True
+    normalize = bool(args[1]) if len(args) > 1 else True
+    bc_o = o.betweenness(fx["n_vertices"], fx["src"], fx["dst"], directed, normalize)
+    bc_r = ref.betweenness(fx["n_vertices"], fx["src"], fx["dst"], directed, normalize,
+                           n_threads=1)
+    assert np.abs(bc_o - bc_r).max() < 1e-9, fx["name"]
+    exp = {row["node_id"]: row[[k for k in row if k != "node_id"][0]]
+           for row in fx["expected"]}
+    for dense, pid in enumerate(fx["node_props"]):
+        if pid in exp:
+            assert abs(bc_o[dense] - exp[pid]) < 1e-3, (fx["name"], pid, bc_o[dense])
 
 
 def check_katz(o, ref, fx):
@@ -174,6 +194,8 @@ def gen_e2e():
                     check_community(o, ref, fx)
                 elif algo == "katz":
                     check_katz(o, ref, fx)
+                elif algo == "betweenness":
+                    check_betweenness(o, ref, fx)
             fixtures.append(fx)
             print(f"ok {fx['name']}: V={fx['n_vertices']} E={len(src)}")
     with open(os.path.join(OUT_DIR, "e2e_cases.json"), "w") as f:

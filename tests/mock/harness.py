@@ -88,7 +88,9 @@ class ModuleHost:
 
     def override_arg(self, pos, value):
         if isinstance(value, bool):
-            raise NotImplementedError
+            self.mock.mock_override_arg_bool(ctypes.c_int64(pos),
+                                             ctypes.c_int64(1 if value else 0))
+            return
         if isinstance(value, int):
             self.mock.mock_override_arg_int(ctypes.c_int64(pos), ctypes.c_int64(value))
         elif isinstance(value, float):

@@ -570,6 +570,13 @@ const char *mock_proc_result_type(const char *proc_name, int64_t i) {
   return nullptr;
 }
 
+void mock_override_arg_bool(int64_t pos, int64_t v) {
+  Value val;
+  val.kind = Value::kBool;
+  val.i = v ? 1 : 0;
+  g_arg_overrides.emplace_back((size_t)pos, val);
+}
+
 void mock_override_arg_int(int64_t pos, int64_t v) {
   Value val;
   val.kind = Value::kInt;

@@ -13,7 +13,7 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), "mock"))
 from harness import MODULES_DIR, ModuleHost  # noqa: E402
 
 ALL_MODULES = ["pagerank", "katz_centrality", "community_detection",
-               "weakly_connected_components"]
+               "weakly_connected_components", "betweenness_centrality"]
 
 
 def require_built():
@@ -83,6 +83,15 @@ def test_community_registration():
         ["min_graph_shrink", "int"], ["community_alg_threshold", "float"],
         ["coloring_alg_threshold", "float"], ["num_of_threads", "int"]]
     assert procs["get_subgraph"]["results"] == [["node", "node"], ["community_id", "int"]]
+
+
+def test_betweenness_registration():
+    procs = registration("betweenness_centrality")
+    # betweenness_centrality_module.cpp:77-90
+    assert procs["get"]["args"] == [
+        ["directed", "bool"], ["normalized", "bool"], ["threads", "int"]]
+    assert procs["get"]["results"] == [
+        ["node", "node"], ["betweenness_centrality", "float"]]
 
 
 FAIL_LOUD_SCRIPT = r"""

@@ -164,3 +164,31 @@ def test_louvain_no_edges(oracle):
     comm, n = oracle.louvain(3, [], [])
     assert n == 0
     assert np.all(comm == -1)
+
+
+# --- betweenness --------------------------------------
+
+@pytest.mark.parametrize("fx", by_algo("betweenness"), ids=lambda fx: fx["name"])
+def test_betweenness_e2e(oracle, fx):
+    args = fx["args"]
+    directed = bool(args[0]) if len(args) > 0 else True
+    normalize = bool(args[1]) if len(args) > 1 else True
+    bc = oracle.betweenness(fx["n_vertices"], fx["src"], fx["dst"], directed, normalize)
+    for row in fx["expected"]:
+        pid = row["node_id"]
+        exp = row[[k for k in row if k != "node_id"][0]]
+        dense = fx["node_props"].index(pid)
+        assert abs(bc[dense] - exp) < 1e-3, (fx["name"], pid)
+
+
+def test_betweenness_vs_reference(oracle, reference):
+    rng = np.random.default_rng(23)
+    for _ in range(6):
+        nv = int(rng.integers(2, 150))
+        ne = int(rng.integers(0, 600))
+        src = rng.integers(0, nv, ne)
+        dst = rng.integers(0, nv, ne)
+        for directed in (True, False):
+            a = oracle.betweenness(nv, src, dst, directed, True)
+            b = reference.betweenness(nv, src, dst, directed, True, n_threads=1)
+            assert np.abs(a - b).max() < 1e-9
