@@ -107,11 +107,12 @@ __device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) 
         if (A.mode == 3) acc += A.partial[row];
         const float newr = (float)((double)A.base_term + (double)A.damping * acc);
         const int64_t gv = A.row_base + row;
-        const float oldr = A.rank_old[gv];
         A.rank_new[gv] = newr;
         A.contrib_new[gv] = newr * A.inv_outdeg[gv];
-        const float d = fabsf(newr - oldr);
-        if (d > maxd) maxd = d;
+        if (A.delta_max) {  // old-rank read only paid when Linf is tracked
+          const float d = fabsf(newr - A.rank_old[gv]);
+          if (d > maxd) maxd = d;
+        }
       }
     }
     if constexpr (LANES > 64) __syncthreads();
