@@ -54,12 +54,6 @@ __global__ void k_weights(int64_t n_edges, uint64_t mixed_seed, float *w) {
   }
 }
 
-__global__ void k_zero_u32(int64_t n, uint32_t *p) {
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x)
-    p[i] = 0;
-}
-
 __global__ void k_hist(int64_t n_edges, const int32_t *idx, uint32_t *counts) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
        i += (int64_t)gridDim.x * blockDim.x)
@@ -81,42 +75,6 @@ __global__ void k_hist_ranged(int64_t n_edges, const int32_t *idx, int32_t lo, i
        i += (int64_t)gridDim.x * blockDim.x) {
     int32_t v = idx[i];
     if (v >= lo && v < hi) atomicAdd(&counts[v - lo], 1u);
-  }
-}
-
-__global__ void k_scatter_in(int64_t n_edges, const int32_t *src, const int32_t *dst,
-                             uint32_t *cursor, int32_t *col) {
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    uint32_t pos = atomicAdd(&cursor[dst[i]], 1u);
-    col[pos] = src[i];
-  }
-}
-
-__global__ void k_scatter_in_ranged(int64_t n_edges, const int32_t *src, const int32_t *dst,
-                                    int32_t lo, int32_t hi, uint32_t *cursor, int32_t *col) {
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    int32_t v = dst[i];
-    if (v >= lo && v < hi) {
-      uint32_t pos = atomicAdd(&cursor[v - lo], 1u);
-      col[pos] = src[i];
-    }
-  }
-}
-
-__global__ void k_scatter_sym(int64_t n_edges, const int32_t *src, const int32_t *dst,
-                              const float *w, uint32_t *cursor, int32_t *col, float *out_w) {
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_edges;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    const int32_t s = src[i], d = dst[i];
-    const float wi = w ? w[i] : 1.0f;
-    uint32_t pos = atomicAdd(&cursor[s], 1u);
-    col[pos] = d;
-    if (out_w) out_w[pos] = wi;
-    pos = atomicAdd(&cursor[d], 1u);
-    col[pos] = s;
-    if (out_w) out_w[pos] = wi;
   }
 }
 

@@ -125,12 +125,6 @@ __global__ void k_fill_f64(int64_t n, double v, double *p) {
     p[i] = v;
 }
 
-__global__ void k_iota_u32(int64_t n, uint32_t *p) {
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x)
-    p[i] = (uint32_t)i;
-}
-
 // Stage the convergence-sort input in ORIGINAL vertex order so the stable
 // descending radix sort breaks centrality ties by original id — the
 // oracle's (documented) tie rule — while values stay permuted indices for

@@ -219,12 +219,6 @@ struct Best {
   int32_t cid;
 };
 
-__device__ inline Best best_merge(Best a, Best b) {
-  if (b.gain > a.gain) return b;
-  if (b.gain == a.gain && b.cid < a.cid) return b;
-  return a;
-}
-
 // One wave processes one small row using its LDS table region.
 __global__ void __launch_bounds__(kBlock) k_sweep_small(SweepArgs A) {
   __shared__ int32_t keys[kBlock / 64][kLdsCap];
