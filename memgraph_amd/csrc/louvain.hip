@@ -78,22 +78,17 @@ __global__ void k_row_wsum_init(int64_t nv, double *vdeg, double *cinfo_deg,
   }
 }
 
-__global__ void __launch_bounds__(kBlock) k_row_wsum_small(
-    int64_t n_small, const int32_t *small_rows, const uint32_t *row_ptr, const double *w,
-    double *vdeg, double *cinfo_deg) {
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int64_t waves_per_grid = (int64_t)gridDim.x * (kBlock / 64);
-  for (int64_t wi = (int64_t)blockIdx.x * (kBlock / 64) + wave; wi < n_small;
-       wi += waves_per_grid) {
-    const int32_t row = small_rows[wi];
+__global__ void k_row_wsum_small(int64_t n_small, const int32_t *small_rows,
+                                 const uint32_t *row_ptr, const double *w, double *vdeg,
+                                 double *cinfo_deg) {
+  // Thread per small row (deg < kSmallRowDeg: bounded loop, no hub tails).
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_small;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t row = small_rows[i];
     double acc = 0.0;
-    for (uint32_t j = row_ptr[row] + lane; j < row_ptr[row + 1]; j += 64) acc += w[j];
-    for (int o = 32; o; o >>= 1) acc += __shfl_down(acc, o, 64);
-    if (lane == 0) {
-      vdeg[row] = acc;
-      cinfo_deg[row] = acc;
-    }
+    for (uint32_t j = row_ptr[row]; j < row_ptr[row + 1]; ++j) acc += w[j];
+    vdeg[row] = acc;
+    cinfo_deg[row] = acc;
   }
 }
 
@@ -452,19 +447,21 @@ __global__ void k_pair_keys(int64_t ne2, const int32_t *col, const uint32_t *row
   }
 }
 
-__global__ void k_row_of_entry(int64_t nv, int64_t ne2, const uint32_t *row_ptr,
-                               int32_t *row_of_entry) {
-  // Per-entry binary search (a thread-per-row fill serialized hub rows:
-  // 113 ms/phase measured at RMAT-24; log2(V) cached reads instead).
-  for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < ne2;
-       j += (int64_t)gridDim.x * blockDim.x) {
-    int64_t lo = 0, hi = nv;  // invariant: row_ptr[lo] <= j < row_ptr[hi]
-    while (lo + 1 < hi) {
-      const int64_t mid = (lo + hi) / 2;
-      if ((uint32_t)j >= row_ptr[mid]) lo = mid;
-      else hi = mid;
-    }
-    row_of_entry[j] = (int32_t)lo;
+__global__ void k_row_of_entry_small(int64_t n_small, const int32_t *rows,
+                                     const uint32_t *row_ptr, int32_t *row_of_entry) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_small;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t row = rows[i];
+    for (uint32_t j = row_ptr[row]; j < row_ptr[row + 1]; ++j) row_of_entry[j] = row;
+  }
+}
+
+__global__ void k_row_of_entry_big(int64_t n_big, const int32_t *rows,
+                                   const uint32_t *row_ptr, int32_t *row_of_entry) {
+  for (int64_t bi = blockIdx.x; bi < n_big; bi += gridDim.x) {
+    const int32_t row = rows[bi];
+    for (uint32_t j = row_ptr[row] + threadIdx.x; j < row_ptr[row + 1]; j += kBlock)
+      row_of_entry[j] = row;
   }
 }
 
@@ -643,9 +640,9 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
                      ctx->stream, nv, vdeg.as<double>(), cinfo_deg.as<double>(),
                      cinfo_size.as<int32_t>());
   if (n_small > 0)
-    hipLaunchKernelGGL(k_row_wsum_small, dim3((uint32_t)grid_for(n_small * 64, 4096)),
-                       dim3(kBlock), 0, ctx->stream, n_small, small_rows.as<int32_t>(),
-                       L.row_ptr, L.w, vdeg.as<double>(), cinfo_deg.as<double>());
+    hipLaunchKernelGGL(k_row_wsum_small, dim3((uint32_t)grid_for(n_small)), dim3(kBlock),
+                       0, ctx->stream, n_small, small_rows.as<int32_t>(), L.row_ptr, L.w,
+                       vdeg.as<double>(), cinfo_deg.as<double>());
   if (n_big > 0)
     hipLaunchKernelGGL(k_row_wsum_big, dim3((uint32_t)(n_big < 4096 ? n_big : 4096)),
                        dim3(kBlock), 0, ctx->stream, n_big, big_rows.as<int32_t>(),
@@ -831,8 +828,35 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   const int64_t ne2 = in.ne2;
   DevBuf row_of_entry, keys, vals, keys_sorted, vals_sorted, u_keys, u_vals, u_count;
   MGX_HIP_TRY(row_of_entry.alloc(ne2 * 4));
-  hipLaunchKernelGGL(k_row_of_entry, dim3((uint32_t)grid_for(ne2)), dim3(kBlock), 0,
-                     ctx->stream, in.nv, ne2, in.row_ptr, row_of_entry.as<int32_t>());
+  {
+    // Binned expansion of the CSR's row ids (hub rows block-parallel).
+    DevBuf counters, srows, brows;
+    MGX_HIP_TRY(counters.alloc(4 * 4));
+    MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
+    hipLaunchKernelGGL(k_classify_rows, dim3((uint32_t)grid_for(in.nv)), dim3(kBlock), 0,
+                       ctx->stream, in.nv, in.row_ptr, counters.as<uint32_t>(),
+                       counters.as<uint32_t>() + 1);
+    uint32_t hc[2] = {0, 0};
+    MGX_HIP_TRY(hipMemcpyAsync(hc, counters.p, 8, hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    MGX_HIP_TRY(srows.alloc((hc[0] > 0 ? hc[0] : 1) * 4));
+    MGX_HIP_TRY(brows.alloc((hc[1] > 0 ? hc[1] : 1) * 4));
+    MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
+    hipLaunchKernelGGL(k_fill_rows, dim3((uint32_t)grid_for(in.nv)), dim3(kBlock), 0,
+                       ctx->stream, in.nv, in.row_ptr, srows.as<int32_t>(),
+                       brows.as<int32_t>(), counters.as<uint32_t>(),
+                       counters.as<uint32_t>() + 1);
+    if (hc[0] > 0)
+      hipLaunchKernelGGL(k_row_of_entry_small, dim3((uint32_t)grid_for(hc[0])),
+                         dim3(kBlock), 0, ctx->stream, (int64_t)hc[0], srows.as<int32_t>(),
+                         in.row_ptr, row_of_entry.as<int32_t>());
+    if (hc[1] > 0)
+      hipLaunchKernelGGL(k_row_of_entry_big,
+                         dim3((uint32_t)(hc[1] < 4096 ? hc[1] : 4096)), dim3(kBlock), 0,
+                         ctx->stream, (int64_t)hc[1], brows.as<int32_t>(), in.row_ptr,
+                         row_of_entry.as<int32_t>());
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  }
   MGX_HIP_TRY(keys.alloc(ne2 * 8));
   MGX_HIP_TRY(vals.alloc(ne2 * 8));
   const uint64_t sentinel = ~0ull;
