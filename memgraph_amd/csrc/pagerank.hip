@@ -258,16 +258,13 @@ mgx_status queue_one_iteration(mgx_pagerank_run *run, bool track_delta) {
   }
 
   // Distributed: every rank owns rows [row_begin,row_end); exchange the new
-  // rank+contrib slices so the next gather sees all sources
-  // (ncclAllGather over xGMI).
+  // contrib slices so the next gather sees all sources (ncclAllGather over
+  // xGMI). The rank vector itself is only needed at finish
+  // (normalize) — gathered once there, halving the per-iteration volume.
   if (run->dist) {
-    const int world = mgx_comm_world(ctx);
     const int64_t shard = run->row_end - run->row_begin;  // equal on all ranks (padded)
     MGX_TRY(mgx_comm_allgather_f32(ctx, A.contrib_new + run->row_begin, A.contrib_new,
                                    (size_t)shard));
-    MGX_TRY(mgx_comm_allgather_f32(ctx, A.rank_new + run->row_begin, A.rank_new,
-                                   (size_t)shard));
-    (void)world;
   }
 
   run->cur = 1 - run->cur;
@@ -315,6 +312,12 @@ mgx_status mgx_pagerank_normalize_download(mgx_pagerank_run *run, double *out_ra
   mgx_context *ctx = run->ctx;
   const int64_t V = run->g->n_vertices;
   if (V == 0) return MGX_OK;
+  if (run->dist && run->iterations > 0) {
+    // Assemble the full rank vector once (deferred from the iterations).
+    const int64_t shard = run->row_end - run->row_begin;
+    MGX_TRY(mgx_comm_allgather_f32(ctx, run->rank[run->cur] + run->row_begin,
+                                   run->rank[run->cur], (size_t)shard));
+  }
   // NormalizeRank (reference pagerank.cpp:157-162): divide by the sum.
   MGX_HIP_TRY(hipMemsetAsync(run->d_scratch, 0, sizeof(double), ctx->stream));
   hipLaunchKernelGGL(k_sum_f32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
