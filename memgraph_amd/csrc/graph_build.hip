@@ -173,8 +173,8 @@ mgx_status build_sorted_cols(mgx_context *ctx, const int32_t *d_src, const int32
                              int64_t col_count) {
   if (n_edges == 0) return MGX_OK;
   uint64_t *keys = nullptr, *keys_out = nullptr;
-  MGX_HIP_TRY(hipMalloc(&keys, n_edges * sizeof(uint64_t)));
-  MGX_HIP_TRY(hipMalloc(&keys_out, n_edges * sizeof(uint64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&keys, n_edges * sizeof(uint64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&keys_out, n_edges * sizeof(uint64_t)));
   if (ranged) {
     hipLaunchKernelGGL(k_pack_pairs_ranged, dim3(grid_for(n_edges)), dim3(kBlock), 0,
                        ctx->stream, n_edges, d_src, d_dst, lo, hi, keys);
@@ -196,8 +196,8 @@ mgx_status build_sorted_cols(mgx_context *ctx, const int32_t *d_src, const int32
   if (err != hipSuccess) return MGX_ERR_HIP;
   hipLaunchKernelGGL(k_unpack_cols, dim3(grid_for(col_count)), dim3(kBlock), 0, ctx->stream,
                      col_count, keys_out, col);
-  MGX_HIP_TRY(hipFree(keys));
-  MGX_HIP_TRY(hipFree(keys_out));
+  MGX_TRY(ctx->free_async(keys));
+  MGX_TRY(ctx->free_async(keys_out));
   return MGX_OK;
 }
 
@@ -226,11 +226,11 @@ mgx_status build_sorted_sym(mgx_context *ctx, const int32_t *d_src, const int32_
   const int64_t n2 = 2 * n_edges;
   uint64_t *keys = nullptr, *keys_out = nullptr;
   float *vals = nullptr, *vals_out = nullptr;
-  MGX_HIP_TRY(hipMalloc(&keys, n2 * sizeof(uint64_t)));
-  MGX_HIP_TRY(hipMalloc(&keys_out, n2 * sizeof(uint64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&keys, n2 * sizeof(uint64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&keys_out, n2 * sizeof(uint64_t)));
   if (out_w) {
-    MGX_HIP_TRY(hipMalloc(&vals, n2 * sizeof(float)));
-    MGX_HIP_TRY(hipMalloc(&vals_out, n2 * sizeof(float)));
+    MGX_TRY(ctx->alloc_async((void **)&vals, n2 * sizeof(float)));
+    MGX_TRY(ctx->alloc_async((void **)&vals_out, n2 * sizeof(float)));
   }
   hipLaunchKernelGGL(k_pack_sym, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
                      n_edges, d_src, d_dst, d_w, keys, vals);
@@ -262,10 +262,10 @@ mgx_status build_sorted_sym(mgx_context *ctx, const int32_t *d_src, const int32_
     MGX_HIP_TRY(hipMemcpyAsync(out_w, vals_out, n2 * sizeof(float),
                                hipMemcpyDeviceToDevice, ctx->stream));
   }
-  MGX_HIP_TRY(hipFree(keys));
-  MGX_HIP_TRY(hipFree(keys_out));
-  if (vals) MGX_HIP_TRY(hipFree(vals));
-  if (vals_out) MGX_HIP_TRY(hipFree(vals_out));
+  MGX_TRY(ctx->free_async(keys));
+  MGX_TRY(ctx->free_async(keys_out));
+  if (vals) MGX_TRY(ctx->free_async(vals));
+  if (vals_out) MGX_TRY(ctx->free_async(vals_out));
   return MGX_OK;
 }
 

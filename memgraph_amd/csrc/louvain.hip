@@ -576,11 +576,18 @@ __global__ void k_fill_rows(int64_t nv, const uint32_t *row_ptr, int32_t *small_
 // ---------- host orchestration -------------------------------------------
 
 struct DevBuf {
+  // Stream-ordered allocation from the cached mempool: per-level multi-GB
+  // hipMalloc/hipFree pairs cost ~2 s of host time per Louvain call at
+  // RMAT-24 (measured); the pool pays it once per process.
+  mgx_context *ctx = nullptr;
   void *p = nullptr;
   ~DevBuf() {
-    if (p) (void)hipFree(p);
+    if (p) (void)hipFreeAsync(p, ctx->stream);
   }
-  hipError_t alloc(size_t bytes) { return hipMalloc(&p, bytes ? bytes : 1); }
+  hipError_t alloc(mgx_context *c, size_t bytes) {
+    ctx = c;
+    return hipMallocAsync(&p, bytes ? bytes : 1, c->stream);
+  }
   template <typename T>
   T *as() {
     return (T *)p;
@@ -599,15 +606,15 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
                          int32_t *C, double *out_mod, int64_t *iters_out) {
   const int64_t nv = L.nv;
   DevBuf vdeg, cinfo_deg, cinfo_size, cupd_deg, cupd_size, past, curr, target, scalars;
-  MGX_HIP_TRY(vdeg.alloc(nv * 8));
-  MGX_HIP_TRY(cinfo_deg.alloc(nv * 8));
-  MGX_HIP_TRY(cinfo_size.alloc(nv * 4));
-  MGX_HIP_TRY(cupd_deg.alloc(nv * 8));
-  MGX_HIP_TRY(cupd_size.alloc(nv * 4));
-  MGX_HIP_TRY(past.alloc(nv * 4));
-  MGX_HIP_TRY(curr.alloc(nv * 4));
-  MGX_HIP_TRY(target.alloc(nv * 4));
-  MGX_HIP_TRY(scalars.alloc(3 * 8));  // [e_xx, a2_x, total_w]
+  MGX_HIP_TRY(vdeg.alloc(ctx, nv * 8));
+  MGX_HIP_TRY(cinfo_deg.alloc(ctx, nv * 8));
+  MGX_HIP_TRY(cinfo_size.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(cupd_deg.alloc(ctx, nv * 8));
+  MGX_HIP_TRY(cupd_size.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(past.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(curr.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(target.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(scalars.alloc(ctx, 3 * 8));  // [e_xx, a2_x, total_w]
 
   MGX_HIP_TRY(hipMemsetAsync(scalars.as<double>() + 2, 0, 8, ctx->stream));
   hipLaunchKernelGGL(k_iota_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
@@ -619,7 +626,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
 
   // Row classification + big-row pool.
   DevBuf counters, small_rows, big_rows, pool_off, pool_keys, pool_vals;
-  MGX_HIP_TRY(counters.alloc(4 * 4));
+  MGX_HIP_TRY(counters.alloc(ctx, 4 * 4));
   MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
   hipLaunchKernelGGL(k_classify_rows, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                      ctx->stream, nv, L.row_ptr, counters.as<uint32_t>(),
@@ -628,8 +635,8 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   MGX_HIP_TRY(hipMemcpyAsync(h_counts, counters.p, 8, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   const int64_t n_small = h_counts[0], n_big = h_counts[1];
-  MGX_HIP_TRY(small_rows.alloc(n_small * 4));
-  MGX_HIP_TRY(big_rows.alloc(n_big * 4));
+  MGX_HIP_TRY(small_rows.alloc(ctx, n_small * 4));
+  MGX_HIP_TRY(big_rows.alloc(ctx, n_big * 4));
   MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
   hipLaunchKernelGGL(k_fill_rows, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                      ctx->stream, nv, L.row_ptr, small_rows.as<int32_t>(),
@@ -656,11 +663,11 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   uint64_t pool_total = 0;
   if (n_big > 0) {
     DevBuf caps;
-    MGX_HIP_TRY(caps.alloc(n_big * 8));
+    MGX_HIP_TRY(caps.alloc(ctx, n_big * 8));
     hipLaunchKernelGGL(k_big_caps, dim3((uint32_t)grid_for(n_big)), dim3(kBlock), 0,
                        ctx->stream, n_big, big_rows.as<int32_t>(), L.row_ptr,
                        caps.as<uint64_t>());
-    MGX_HIP_TRY(pool_off.alloc((n_big + 1) * 8));
+    MGX_HIP_TRY(pool_off.alloc(ctx, (n_big + 1) * 8));
     size_t tmp_bytes = 0;
     auto err = rocprim::exclusive_scan(nullptr, tmp_bytes, caps.as<uint64_t>(),
                                        pool_off.as<uint64_t>(), (uint64_t)0, n_big,
@@ -681,8 +688,8 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     pool_total = last_off + last_cap;
     MGX_HIP_TRY(hipMemcpyAsync(pool_off.as<uint64_t>() + n_big, &pool_total, 8,
                                hipMemcpyHostToDevice, ctx->stream));
-    MGX_HIP_TRY(pool_keys.alloc(pool_total * 4));
-    MGX_HIP_TRY(pool_vals.alloc(pool_total * 8));
+    MGX_HIP_TRY(pool_keys.alloc(ctx, pool_total * 4));
+    MGX_HIP_TRY(pool_vals.alloc(ctx, pool_total * 8));
   }
 
   SweepArgs A;
@@ -757,15 +764,15 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
 // renumberClustersContiguously: new id = rank of community's min vertex.
 mgx_status renumber(mgx_context *ctx, int32_t *C, int64_t nv, int64_t *n_clusters) {
   DevBuf rep, keys_out, cand, cand_out, newid, nact;
-  MGX_HIP_TRY(rep.alloc(nv * 4));
+  MGX_HIP_TRY(rep.alloc(ctx, nv * 4));
   hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
                      nv, INT32_MAX, rep.as<int32_t>());
   hipLaunchKernelGGL(k_rep_min, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
                      nv, C, rep.as<int32_t>());
   // sort (rep, c) ascending by rep; INT32_MAX reps (inactive ids) sort last.
-  MGX_HIP_TRY(cand.alloc(nv * 4));
-  MGX_HIP_TRY(cand_out.alloc(nv * 4));
-  MGX_HIP_TRY(keys_out.alloc(nv * 4));
+  MGX_HIP_TRY(cand.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(cand_out.alloc(ctx, nv * 4));
+  MGX_HIP_TRY(keys_out.alloc(ctx, nv * 4));
   hipLaunchKernelGGL(k_iota_i32, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0, ctx->stream,
                      nv, cand.as<int32_t>());
   size_t tmp_bytes = 0;
@@ -782,7 +789,7 @@ mgx_status renumber(mgx_context *ctx, int32_t *C, int64_t nv, int64_t *n_cluster
   // count active = reps != INT32_MAX: binary property; count via reduce on
   // host copy of the boundary — simpler: count nonzero with a kernel.
   DevBuf count;
-  MGX_HIP_TRY(count.alloc(8));
+  MGX_HIP_TRY(count.alloc(ctx, 8));
   MGX_HIP_TRY(hipMemsetAsync(count.p, 0, 8, ctx->stream));
   {
     struct K {
@@ -810,7 +817,7 @@ mgx_status renumber(mgx_context *ctx, int32_t *C, int64_t nv, int64_t *n_cluster
   unsigned long long n_active = 0;
   MGX_HIP_TRY(hipMemcpyAsync(&n_active, count.p, 8, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  MGX_HIP_TRY(newid.alloc(nv * 4));
+  MGX_HIP_TRY(newid.alloc(ctx, nv * 4));
   hipLaunchKernelGGL(k_newid_from_sorted, dim3((uint32_t)grid_for((int64_t)n_active)),
                      dim3(kBlock), 0, ctx->stream, (int64_t)n_active,
                      cand_out.as<int32_t>(), newid.as<int32_t>());
@@ -827,11 +834,11 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
                    Level *out) {
   const int64_t ne2 = in.ne2;
   DevBuf row_of_entry, keys, vals, keys_sorted, vals_sorted, u_keys, u_vals, u_count;
-  MGX_HIP_TRY(row_of_entry.alloc(ne2 * 4));
+  MGX_HIP_TRY(row_of_entry.alloc(ctx, ne2 * 4));
   {
     // Binned expansion of the CSR's row ids (hub rows block-parallel).
     DevBuf counters, srows, brows;
-    MGX_HIP_TRY(counters.alloc(4 * 4));
+    MGX_HIP_TRY(counters.alloc(ctx, 4 * 4));
     MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
     hipLaunchKernelGGL(k_classify_rows, dim3((uint32_t)grid_for(in.nv)), dim3(kBlock), 0,
                        ctx->stream, in.nv, in.row_ptr, counters.as<uint32_t>(),
@@ -839,8 +846,8 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
     uint32_t hc[2] = {0, 0};
     MGX_HIP_TRY(hipMemcpyAsync(hc, counters.p, 8, hipMemcpyDeviceToHost, ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-    MGX_HIP_TRY(srows.alloc((hc[0] > 0 ? hc[0] : 1) * 4));
-    MGX_HIP_TRY(brows.alloc((hc[1] > 0 ? hc[1] : 1) * 4));
+    MGX_HIP_TRY(srows.alloc(ctx, (hc[0] > 0 ? hc[0] : 1) * 4));
+    MGX_HIP_TRY(brows.alloc(ctx, (hc[1] > 0 ? hc[1] : 1) * 4));
     MGX_HIP_TRY(hipMemsetAsync(counters.p, 0, 16, ctx->stream));
     hipLaunchKernelGGL(k_fill_rows, dim3((uint32_t)grid_for(in.nv)), dim3(kBlock), 0,
                        ctx->stream, in.nv, in.row_ptr, srows.as<int32_t>(),
@@ -857,15 +864,15 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
                          row_of_entry.as<int32_t>());
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   }
-  MGX_HIP_TRY(keys.alloc(ne2 * 8));
-  MGX_HIP_TRY(vals.alloc(ne2 * 8));
+  MGX_HIP_TRY(keys.alloc(ctx, ne2 * 8));
+  MGX_HIP_TRY(vals.alloc(ctx, ne2 * 8));
   const uint64_t sentinel = ~0ull;
   hipLaunchKernelGGL(k_pair_keys, dim3((uint32_t)grid_for(ne2)), dim3(kBlock), 0,
                      ctx->stream, ne2, in.col, in.row_ptr, in.nv, C, in.w,
                      row_of_entry.as<int32_t>(), keys.as<uint64_t>(), vals.as<double>(),
                      sentinel);
-  MGX_HIP_TRY(keys_sorted.alloc(ne2 * 8));
-  MGX_HIP_TRY(vals_sorted.alloc(ne2 * 8));
+  MGX_HIP_TRY(keys_sorted.alloc(ctx, ne2 * 8));
+  MGX_HIP_TRY(vals_sorted.alloc(ctx, ne2 * 8));
   size_t tmp_bytes = 0;
   auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys.as<uint64_t>(),
                                        keys_sorted.as<uint64_t>(), vals.as<double>(),
@@ -877,9 +884,9 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
                                   keys_sorted.as<uint64_t>(), vals.as<double>(),
                                   vals_sorted.as<double>(), ne2, 0, 64, ctx->stream);
   if (err != hipSuccess) return MGX_ERR_HIP;
-  MGX_HIP_TRY(u_keys.alloc(ne2 * 8));
-  MGX_HIP_TRY(u_vals.alloc(ne2 * 8));
-  MGX_HIP_TRY(u_count.alloc(8));
+  MGX_HIP_TRY(u_keys.alloc(ctx, ne2 * 8));
+  MGX_HIP_TRY(u_vals.alloc(ctx, ne2 * 8));
+  MGX_HIP_TRY(u_count.alloc(ctx, 8));
   err = rocprim::reduce_by_key(nullptr, tmp_bytes, keys_sorted.as<uint64_t>(),
                                vals_sorted.as<double>(), ne2, u_keys.as<uint64_t>(),
                                u_vals.as<double>(), (unsigned int *)u_count.p,
@@ -908,8 +915,8 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
 
   // CSR counts + self presence.
   DevBuf counts, self_present;
-  MGX_HIP_TRY(counts.alloc(n_clusters * 4));
-  MGX_HIP_TRY(self_present.alloc(n_clusters * 4));
+  MGX_HIP_TRY(counts.alloc(ctx, n_clusters * 4));
+  MGX_HIP_TRY(self_present.alloc(ctx, n_clusters * 4));
   MGX_HIP_TRY(hipMemsetAsync(counts.p, 0, n_clusters * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(self_present.p, 0, n_clusters * 4, ctx->stream));
   if (n_pairs > 0)
@@ -1003,7 +1010,7 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
   L.col = g->sym_col;
   L.ne2 = 2 * g->n_edges;
   DevBuf w0;
-  MGX_HIP_TRY(w0.alloc(L.ne2 * 8));
+  MGX_HIP_TRY(w0.alloc(ctx, L.ne2 * 8));
   if (g->sym_w) {
     hipLaunchKernelGGL(k_f32_to_f64, dim3((uint32_t)grid_for(L.ne2)), dim3(kBlock), 0,
                        ctx->stream, L.ne2, g->sym_w, (double *)w0.p);
@@ -1022,7 +1029,7 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
   bool own_level = false;
 
   DevBuf c_orig_d, c_level;
-  MGX_HIP_TRY(c_orig_d.alloc(nv0 * 4));
+  MGX_HIP_TRY(c_orig_d.alloc(ctx, nv0 * 4));
   hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for(nv0)), dim3(kBlock), 0,
                      ctx->stream, nv0, -1, c_orig_d.as<int32_t>());
 
@@ -1031,7 +1038,7 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
   mgx_status status = MGX_OK;
   while (true) {
     DevBuf C;
-    if (C.alloc(L.nv * 4) != hipSuccess) {
+    if (C.alloc(ctx, L.nv * 4) != hipSuccess) {
       status = MGX_ERR_OUT_OF_MEMORY;
       break;
     }
@@ -1068,7 +1075,7 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
 
   // Download.
   DevBuf wide;
-  MGX_HIP_TRY(wide.alloc(nv0 * 8));
+  MGX_HIP_TRY(wide.alloc(ctx, nv0 * 8));
   hipLaunchKernelGGL(k_widen_i32_to_i64, dim3((uint32_t)grid_for(nv0)), dim3(kBlock), 0,
                      ctx->stream, nv0, c_orig_d.as<int32_t>(), wide.as<int64_t>());
   if (out_community) {

@@ -40,6 +40,16 @@ extern "C" int mgx_device_count(void) {
   return n;
 }
 
+mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
+  MGX_HIP_TRY(hipMallocAsync(p, bytes ? bytes : 1, stream));
+  return MGX_OK;
+}
+
+mgx_status mgx_context::free_async(void *p) {
+  if (p) MGX_HIP_TRY(hipFreeAsync(p, stream));
+  return MGX_OK;
+}
+
 mgx_status mgx_context::reserve(size_t bytes, void **out) {
   if (bytes > workspace_bytes) {
     if (workspace) MGX_HIP_TRY(hipFree(workspace));
@@ -71,6 +81,12 @@ extern "C" mgx_status mgx_init(int device, mgx_context **out) {
     delete ctx;
     mgx_set_error("hipStreamCreate failed");
     return MGX_ERR_HIP;
+  }
+  // Cache freed stream-ordered allocations (see mgx_context::alloc_async).
+  hipMemPool_t pool = nullptr;
+  if (hipDeviceGetDefaultMemPool(&pool, device) == hipSuccess && pool) {
+    uint64_t threshold = UINT64_MAX;
+    (void)hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &threshold);
   }
   *out = ctx;
   return MGX_OK;
@@ -154,8 +170,8 @@ mgx_status build_generated(mgx_context *ctx, int64_t n_vertices, int64_t n_edges
   MGX_HIP_TRY(hipSetDevice(ctx->device));
   int32_t *d_src = nullptr, *d_dst = nullptr;
   float *d_w = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(hipMalloc(&d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&d_src, n_edges * sizeof(int32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&d_dst, n_edges * sizeof(int32_t)));
   mgx_status s;
   if (rmat) {
     s = mgx_gen_rmat_device(ctx, scale, n_edges, seed, a, b, c, d_src, d_dst);
@@ -163,7 +179,7 @@ mgx_status build_generated(mgx_context *ctx, int64_t n_vertices, int64_t n_edges
     s = mgx_gen_uniform_device(ctx, n_vertices, n_edges, seed, d_src, d_dst);
   }
   if (s == MGX_OK && (flags & MGX_BUILD_WEIGHTED)) {
-    MGX_HIP_TRY(hipMalloc(&d_w, (n_edges > 0 ? n_edges : 1) * sizeof(float)));
+    MGX_TRY(ctx->alloc_async((void **)&d_w, n_edges * sizeof(float)));
     s = mgx_gen_weights_device(ctx, n_edges, weight_seed, d_w);
   }
   mgx_graph *g = nullptr;
@@ -171,9 +187,9 @@ mgx_status build_generated(mgx_context *ctx, int64_t n_vertices, int64_t n_edges
     g = new mgx_graph();
     s = mgx_build_from_device_coo(ctx, d_src, d_dst, d_w, n_vertices, n_edges, flags, g);
   }
-  (void)hipFree(d_src);
-  (void)hipFree(d_dst);
-  if (d_w) (void)hipFree(d_w);
+  (void)ctx->free_async(d_src);
+  (void)ctx->free_async(d_dst);
+  if (d_w) (void)ctx->free_async(d_w);
   if (s != MGX_OK) {
     if (g) (void)mgx_graph_destroy(ctx, g);
     return s;

@@ -35,6 +35,12 @@ struct mgx_comm_state;  // comm.cpp (RCCL)
 struct mgx_context {
   int device = -1;
   hipStream_t stream = nullptr;
+  // Stream-ordered allocations for large transient buffers (sort keys, COO
+  // staging): the default mempool with an unbounded release threshold keeps
+  // freed blocks cached, so repeated multi-GB hipMalloc page-table setup
+  // (measured up to ~1 s per RMAT-26 build) is paid once per process.
+  mgx_status alloc_async(void **p, size_t bytes);
+  mgx_status free_async(void *p);
   // Grow-only device workspace for rocPRIM temp storage etc.
   void *workspace = nullptr;
   size_t workspace_bytes = 0;
