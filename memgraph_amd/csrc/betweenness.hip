@@ -143,11 +143,11 @@ mgx_status mgx_betweenness_impl(mgx_context *ctx, mgx_graph *g, int directed, in
   unsigned long long *sigma = nullptr;
   double *dep = nullptr, *bc = nullptr;
   uint32_t *d_changed = nullptr;
-  MGX_HIP_TRY(hipMalloc(&dist, batch * V * sizeof(int32_t)));
-  MGX_HIP_TRY(hipMalloc(&sigma, batch * V * sizeof(unsigned long long)));
-  MGX_HIP_TRY(hipMalloc(&dep, batch * V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&bc, V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&d_changed, 4));
+  MGX_TRY(ctx->alloc_async((void **)&dist, batch * V * sizeof(int32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&sigma, batch * V * sizeof(unsigned long long)));
+  MGX_TRY(ctx->alloc_async((void **)&dep, batch * V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&bc, V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&d_changed, 4));
   MGX_HIP_TRY(hipMemsetAsync(bc, 0, V * sizeof(double), ctx->stream));
 
   for (int64_t s0 = 0; s0 < V; s0 += batch) {
@@ -197,11 +197,11 @@ mgx_status mgx_betweenness_impl(mgx_context *ctx, mgx_graph *g, int directed, in
                                ctx->stream));
   }
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  MGX_HIP_TRY(hipFree(dist));
-  MGX_HIP_TRY(hipFree(sigma));
-  MGX_HIP_TRY(hipFree(dep));
-  MGX_HIP_TRY(hipFree(bc));
-  MGX_HIP_TRY(hipFree(d_changed));
+  MGX_TRY(ctx->free_async(dist));
+  MGX_TRY(ctx->free_async(sigma));
+  MGX_TRY(ctx->free_async(dep));
+  MGX_TRY(ctx->free_async(bc));
+  MGX_TRY(ctx->free_async(d_changed));
   MGX_HIP_TRY(hipGetLastError());
   return MGX_OK;
 }

@@ -184,22 +184,22 @@ mgx_status mgx_katz_impl(mgx_context *ctx, mgx_graph *g, double alpha, double ep
 
   // MaxDegree over OUT-degrees (katz.cpp:137-148 / mg_graph.hpp:96-109).
   uint32_t *d_max = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_max, 4));
+  MGX_TRY(ctx->alloc_async((void **)&d_max, 4));
   MGX_HIP_TRY(hipMemsetAsync(d_max, 0, 4, ctx->stream));
   hipLaunchKernelGGL(k_max_u32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
                      V, g->out_degree, d_max);
   uint32_t deg_max = 0;
   MGX_HIP_TRY(hipMemcpyAsync(&deg_max, d_max, 4, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  MGX_HIP_TRY(hipFree(d_max));
+  MGX_TRY(ctx->free_async(d_max));
   const double gamma = (double)deg_max / (1.0 - alpha * alpha * (double)deg_max);
 
   double *omega[2] = {nullptr, nullptr}, *cent = nullptr, *lr = nullptr, *ur = nullptr;
-  MGX_HIP_TRY(hipMalloc(&omega[0], V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&omega[1], V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&cent, V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&lr, V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&ur, V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&omega[0], V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&omega[1], V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&cent, V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&lr, V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&ur, V * sizeof(double)));
   hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
                      V, 1.0, omega[0]);  // omega_0 = 1 (katz.cpp:41-44)
   MGX_HIP_TRY(hipMemsetAsync(cent, 0, V * sizeof(double), ctx->stream));
@@ -209,11 +209,11 @@ mgx_status mgx_katz_impl(mgx_context *ctx, mgx_graph *g, double alpha, double ep
   // Sort buffers for the convergence test.
   double *keys_in = nullptr, *keys_out = nullptr;
   uint32_t *vals_in = nullptr, *vals_out = nullptr, *d_flag = nullptr;
-  MGX_HIP_TRY(hipMalloc(&keys_in, V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&keys_out, V * sizeof(double)));
-  MGX_HIP_TRY(hipMalloc(&vals_in, V * sizeof(uint32_t)));
-  MGX_HIP_TRY(hipMalloc(&vals_out, V * sizeof(uint32_t)));
-  MGX_HIP_TRY(hipMalloc(&d_flag, 4));
+  MGX_TRY(ctx->alloc_async((void **)&keys_in, V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&keys_out, V * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&vals_in, V * sizeof(uint32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&vals_out, V * sizeof(uint32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&d_flag, 4));
 
   KatzArgs A;
   A.row_ptr = g->in_row_ptr;
@@ -286,16 +286,16 @@ mgx_status mgx_katz_impl(mgx_context *ctx, mgx_graph *g, double alpha, double ep
   }
   if (iterations) *iterations = iter;
 
-  MGX_HIP_TRY(hipFree(omega[0]));
-  MGX_HIP_TRY(hipFree(omega[1]));
-  MGX_HIP_TRY(hipFree(cent));
-  MGX_HIP_TRY(hipFree(lr));
-  MGX_HIP_TRY(hipFree(ur));
-  MGX_HIP_TRY(hipFree(keys_in));
-  MGX_HIP_TRY(hipFree(keys_out));
-  MGX_HIP_TRY(hipFree(vals_in));
-  MGX_HIP_TRY(hipFree(vals_out));
-  MGX_HIP_TRY(hipFree(d_flag));
+  MGX_TRY(ctx->free_async(omega[0]));
+  MGX_TRY(ctx->free_async(omega[1]));
+  MGX_TRY(ctx->free_async(cent));
+  MGX_TRY(ctx->free_async(lr));
+  MGX_TRY(ctx->free_async(ur));
+  MGX_TRY(ctx->free_async(keys_in));
+  MGX_TRY(ctx->free_async(keys_out));
+  MGX_TRY(ctx->free_async(vals_in));
+  MGX_TRY(ctx->free_async(vals_out));
+  MGX_TRY(ctx->free_async(d_flag));
   MGX_HIP_TRY(hipGetLastError());
   return status;
 }

@@ -354,14 +354,14 @@ mgx_status pagerank_start_common(mgx_context *ctx, mgx_graph *g, double damping,
   const int64_t alloc = dist ? (row_end - row_begin) * (int64_t)mgx_comm_world(ctx) : V;
   const int64_t n = (alloc > V ? alloc : V);
   for (int i = 0; i < 2; ++i) {
-    MGX_HIP_TRY(hipMalloc(&run->rank[i], (n > 0 ? n : 1) * sizeof(float)));
-    MGX_HIP_TRY(hipMalloc(&run->contrib[i], (n > 0 ? n : 1) * sizeof(float)));
+    MGX_TRY(ctx->alloc_async((void **)&run->rank[i], n * sizeof(float)));
+    MGX_TRY(ctx->alloc_async((void **)&run->contrib[i], n * sizeof(float)));
   }
-  MGX_HIP_TRY(hipMalloc(&run->d_delta, sizeof(uint32_t)));
-  MGX_HIP_TRY(hipMalloc(&run->d_scratch, (V + 1) * sizeof(double)));
+  MGX_TRY(ctx->alloc_async((void **)&run->d_delta, sizeof(uint32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&run->d_scratch, (V + 1) * sizeof(double)));
   if (g->n_stripes > 1) {
     const int64_t rows = (dist ? (row_end < V ? row_end : V) : V) - row_begin;
-    MGX_HIP_TRY(hipMalloc(&run->d_partial, (rows > 0 ? rows : 1) * sizeof(double)));
+    MGX_TRY(ctx->alloc_async((void **)&run->d_partial, rows * sizeof(double)));
   }
   if (V > 0) {
     const float r0 = (float)(1.0 / (double)V);
@@ -375,12 +375,12 @@ mgx_status pagerank_start_common(mgx_context *ctx, mgx_graph *g, double damping,
 
 void pagerank_run_free(mgx_pagerank_run *run) {
   for (int i = 0; i < 2; ++i) {
-    if (run->rank[i]) (void)hipFree(run->rank[i]);
-    if (run->contrib[i]) (void)hipFree(run->contrib[i]);
+    if (run->rank[i]) (void)run->ctx->free_async(run->rank[i]);
+    if (run->contrib[i]) (void)run->ctx->free_async(run->contrib[i]);
   }
-  if (run->d_delta) (void)hipFree(run->d_delta);
-  if (run->d_scratch) (void)hipFree(run->d_scratch);
-  if (run->d_partial) (void)hipFree(run->d_partial);
+  if (run->d_delta) (void)run->ctx->free_async(run->d_delta);
+  if (run->d_scratch) (void)run->ctx->free_async(run->d_scratch);
+  if (run->d_partial) (void)run->ctx->free_async(run->d_partial);
   for (auto e : run->ev_start) (void)hipEventDestroy(e);
   for (auto e : run->ev_stop) (void)hipEventDestroy(e);
   delete run;

@@ -159,8 +159,8 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
 
   int32_t *label = nullptr;
   uint32_t *d_changed = nullptr;
-  MGX_HIP_TRY(hipMalloc(&label, V * sizeof(int32_t)));
-  MGX_HIP_TRY(hipMalloc(&d_changed, sizeof(uint32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&label, V * sizeof(int32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&d_changed, sizeof(uint32_t)));
   hipLaunchKernelGGL(k_iota, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                      label);
 
@@ -194,8 +194,8 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
 
   // Renumber + count.
   uint32_t *flag = nullptr, *scan = nullptr;
-  MGX_HIP_TRY(hipMalloc(&flag, V * sizeof(uint32_t)));
-  MGX_HIP_TRY(hipMalloc(&scan, (V + 1) * sizeof(uint32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&flag, V * sizeof(uint32_t)));
+  MGX_TRY(ctx->alloc_async((void **)&scan, (V + 1) * sizeof(uint32_t)));
   hipLaunchKernelGGL(k_wcc_flag_roots, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0,
                      ctx->stream, V, label, flag);
   size_t tmp_bytes = 0;
@@ -209,7 +209,7 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
   if (err != hipSuccess) return MGX_ERR_HIP;
 
   int64_t *d_out = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_out, V * sizeof(int64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&d_out, V * sizeof(int64_t)));
   hipLaunchKernelGGL(k_wcc_emit, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
                      V, label, scan, d_out);
   if (out_component) {
@@ -226,11 +226,11 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
     *n_components = (int64_t)last_scan + last_flag;
   }
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  MGX_HIP_TRY(hipFree(label));
-  MGX_HIP_TRY(hipFree(d_changed));
-  MGX_HIP_TRY(hipFree(flag));
-  MGX_HIP_TRY(hipFree(scan));
-  MGX_HIP_TRY(hipFree(d_out));
+  MGX_TRY(ctx->free_async(label));
+  MGX_TRY(ctx->free_async(d_changed));
+  MGX_TRY(ctx->free_async(flag));
+  MGX_TRY(ctx->free_async(scan));
+  MGX_TRY(ctx->free_async(d_out));
   MGX_HIP_TRY(hipGetLastError());
   return MGX_OK;
 }
