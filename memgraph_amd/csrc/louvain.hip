@@ -43,6 +43,26 @@ constexpr int kBlock = 256;
 constexpr uint32_t kSmallRowDeg = 256;  // <: wave+LDS table; >=: block+pool
 constexpr int kLdsCap = 512;            // per-wave table entries (>= 2*255)
 
+inline bool louvain_debug() {
+  static const bool v = [] {
+    const char *e = getenv("MGX_LOUVAIN_DEBUG");
+    return e && atoi(e) != 0;
+  }();
+  return v;
+}
+
+#define MGX_LDBG(ctx, ...)                                                    \
+  do {                                                                        \
+    if (louvain_debug()) {                                                    \
+      hipError_t _se = hipStreamSynchronize((ctx)->stream);                   \
+      hipError_t _ke = hipGetLastError();                                     \
+      fprintf(stderr, "[louvain] " __VA_ARGS__);                              \
+      fprintf(stderr, " sync=%s last=%s\n", hipGetErrorString(_se),          \
+              hipGetErrorString(_ke));                                        \
+      fflush(stderr);                                                         \
+    }                                                                         \
+  } while (0)
+
 inline int64_t grid_for(int64_t work, int64_t cap = 4096) {
   int64_t g = (work + kBlock - 1) / kBlock;
   if (g < 1) g = 1;
@@ -660,6 +680,8 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   MGX_TRY(read_scalar_f64(ctx, scalars.as<double>() + 2, &total_w));
   const double constant = 1.0 / total_w;  // calConstantForSecondTerm
 
+  MGX_LDBG(ctx, "level nv=%lld n_small=%lld n_big=%lld wsum done", (long long)nv,
+           (long long)n_small, (long long)n_big);
   uint64_t pool_total = 0;
   if (n_big > 0) {
     DevBuf caps;
@@ -692,6 +714,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     MGX_HIP_TRY(pool_vals.alloc(ctx, pool_total * 8));
   }
 
+  MGX_LDBG(ctx, "pool_total=%llu", (unsigned long long)pool_total);
   SweepArgs A;
   A.row_ptr = L.row_ptr;
   A.col = L.col;
@@ -720,6 +743,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     MGX_HIP_TRY(hipMemsetAsync(scalars.p, 0, 16, ctx->stream));  // e_xx, a2_x
     A.curr = p_curr;
     A.target = p_target;
+    MGX_LDBG(ctx, "iter=%lld pre-sweep", (long long)iters);
     hipLaunchKernelGGL(k_sweep_empty, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, L.row_ptr, p_target);
     if (n_small > 0) {
@@ -733,6 +757,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
       hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
                          A);
     }
+    MGX_LDBG(ctx, "iter=%lld post-sweep", (long long)iters);
     hipLaunchKernelGGL(k_sum_sq_f64, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, cinfo_deg.as<double>(), scalars.as<double>() + 1);
     double exx_a2x[2] = {0.0, 0.0};
@@ -1044,6 +1069,8 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
     }
     prev_mod = curr_mod;
     int64_t iters = 0;
+    MGX_LDBG(ctx, "phase=%lld level nv=%lld ne2=%lld", (long long)phase, (long long)L.nv,
+             (long long)L.ne2);
     status = louvain_level(ctx, L, prev_mod, threshold, C.as<int32_t>(), &curr_mod, &iters);
     if (status != MGX_OK) break;
     tot_itr += iters;
@@ -1060,6 +1087,8 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
     if (phase > 200 || tot_itr > 100000) break;
     if ((curr_mod - prev_mod) > threshold) {
       Level next;
+      MGX_LDBG(ctx, "coarsen from nv=%lld to %lld", (long long)L.nv,
+               (long long)n_clusters);
       status = coarsen(ctx, L, C.as<int32_t>(), n_clusters, &next);
       free_level(&L, own_level);
       if (status != MGX_OK) break;

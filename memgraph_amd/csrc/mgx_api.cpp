@@ -40,13 +40,32 @@ extern "C" int mgx_device_count(void) {
   return n;
 }
 
+namespace {
+bool no_async_pool() {
+  static const bool v = [] {
+    const char *e = getenv("MGX_NO_ASYNC_POOL");
+    return e && atoi(e) != 0;
+  }();
+  return v;
+}
+}  // namespace
+
 mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
+  if (no_async_pool()) {
+    MGX_HIP_TRY(hipMalloc(p, bytes ? bytes : 1));
+    return MGX_OK;
+  }
   MGX_HIP_TRY(hipMallocAsync(p, bytes ? bytes : 1, stream));
   return MGX_OK;
 }
 
 mgx_status mgx_context::free_async(void *p) {
-  if (p) MGX_HIP_TRY(hipFreeAsync(p, stream));
+  if (!p) return MGX_OK;
+  if (no_async_pool()) {
+    MGX_HIP_TRY(hipFree(p));
+    return MGX_OK;
+  }
+  MGX_HIP_TRY(hipFreeAsync(p, stream));
   return MGX_OK;
 }
 
