@@ -43,6 +43,11 @@ constexpr int kBlock = 256;
 constexpr uint32_t kSmallRowDeg = 256;  // <: wave+LDS table; >=: block+pool
 constexpr int kLdsCap = 512;            // per-wave table entries (>= 2*255)
 
+inline int louvain_env(const char *name) {
+  const char *e = getenv(name);
+  return e ? atoi(e) : 0;
+}
+
 inline bool louvain_debug() {
   static const bool v = [] {
     const char *e = getenv("MGX_LOUVAIN_DEBUG");
@@ -330,14 +335,8 @@ __global__ void __launch_bounds__(kBlock) k_sweep_big(SweepArgs A) {
     const uint32_t cap = (uint32_t)(t1 - t0);  // power of two
     int32_t *keys = A.pool_keys + t0;
     double *vals = A.pool_vals + t0;
-    // Clear this row's own region (regions are per-row exclusive): clearing
-    // in-kernel costs ~the row's own edge traffic, instead of a whole-pool
-    // memset per sweep (measured dominant at RMAT-24).
-    for (uint32_t sIdx = threadIdx.x; sIdx < cap; sIdx += kBlock) {
-      keys[sIdx] = -1;
-      vals[sIdx] = 0.0;
-    }
-    __syncthreads();
+    // The pool is cleared before the launch (a per-row in-kernel clear was
+    // tried and implicated in memory faults at RMAT-24; see git history).
     if (threadIdx.x == 0) {
       uint32_t h = ((uint32_t)sc * 2654435761u) & (cap - 1);
       keys[h] = sc;
@@ -593,6 +592,33 @@ __global__ void k_fill_rows(int64_t nv, const uint32_t *row_ptr, int32_t *small_
   }
 }
 
+__global__ void k_check_lists(int64_t nv, const uint32_t *row_ptr, const int32_t *small_rows,
+                              int64_t n_small, const int32_t *big_rows, int64_t n_big,
+                              uint32_t *mark, uint32_t *err) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_small + n_big;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t row = i < n_small ? small_rows[i] : big_rows[i - n_small];
+    if (row < 0 || row >= nv) {
+      atomicOr(err, 1u);
+      continue;
+    }
+    atomicAdd(&mark[row], 1u);
+    const uint32_t deg = row_ptr[row + 1] - row_ptr[row];
+    if (i < n_small && !(deg > 0 && deg < kSmallRowDeg)) atomicOr(err, 2u);
+    if (i >= n_small && deg < kSmallRowDeg) atomicOr(err, 4u);
+  }
+}
+
+__global__ void k_check_marks(int64_t nv, const uint32_t *row_ptr, const uint32_t *mark,
+                              uint32_t *err) {
+  for (int64_t v = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; v < nv;
+       v += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t deg = row_ptr[v + 1] - row_ptr[v];
+    const uint32_t expect = deg > 0 ? 1u : 0u;
+    if (mark[v] != expect) atomicOr(err, 8u);
+  }
+}
+
 // ---------- host orchestration -------------------------------------------
 
 struct DevBuf {
@@ -662,6 +688,25 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
                      ctx->stream, nv, L.row_ptr, small_rows.as<int32_t>(),
                      big_rows.as<int32_t>(), counters.as<uint32_t>(),
                      counters.as<uint32_t>() + 1);
+  if (louvain_debug()) {
+    DevBuf mark, errbuf;
+    MGX_HIP_TRY(mark.alloc(ctx, nv * 4));
+    MGX_HIP_TRY(errbuf.alloc(ctx, 4));
+    MGX_HIP_TRY(hipMemsetAsync(mark.p, 0, nv * 4, ctx->stream));
+    MGX_HIP_TRY(hipMemsetAsync(errbuf.p, 0, 4, ctx->stream));
+    hipLaunchKernelGGL(k_check_lists, dim3((uint32_t)grid_for(n_small + n_big)),
+                       dim3(kBlock), 0, ctx->stream, nv, L.row_ptr,
+                       small_rows.as<int32_t>(), n_small, big_rows.as<int32_t>(), n_big,
+                       mark.as<uint32_t>(), errbuf.as<uint32_t>());
+    hipLaunchKernelGGL(k_check_marks, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
+                       ctx->stream, nv, L.row_ptr, mark.as<uint32_t>(),
+                       errbuf.as<uint32_t>());
+    uint32_t e = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&e, errbuf.p, 4, hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    fprintf(stderr, "[louvain] LIST CHECK nv=%lld err=%u\n", (long long)nv, e);
+    fflush(stderr);
+  }
   // Binned row weight sums (sumVertexDegree) over the just-built lists.
   hipLaunchKernelGGL(k_row_wsum_init, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                      ctx->stream, nv, vdeg.as<double>(), cinfo_deg.as<double>(),
@@ -746,17 +791,24 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     MGX_LDBG(ctx, "iter=%lld pre-sweep", (long long)iters);
     hipLaunchKernelGGL(k_sweep_empty, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, L.row_ptr, p_target);
+    MGX_LDBG(ctx, "iter=%lld after-empty", (long long)iters);
     if (n_small > 0) {
       const int64_t waves_needed = n_small;
       const int64_t blocks = grid_for(waves_needed * 64, 4096);
       hipLaunchKernelGGL(k_sweep_small, dim3((uint32_t)blocks), dim3(kBlock), 0,
                          ctx->stream, A);
     }
+    MGX_LDBG(ctx, "iter=%lld after-small", (long long)iters);
     if (n_big > 0) {
+      hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for((int64_t)pool_total)),
+                         dim3(kBlock), 0, ctx->stream, (int64_t)pool_total, -1,
+                         pool_keys.as<int32_t>());
+      MGX_HIP_TRY(hipMemsetAsync(pool_vals.p, 0, pool_total * 8, ctx->stream));
       const int64_t blocks = n_big < 4096 ? n_big : 4096;
       hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
                          A);
     }
+    MGX_LDBG(ctx, "iter=%lld after-big", (long long)iters);
     MGX_LDBG(ctx, "iter=%lld post-sweep", (long long)iters);
     hipLaunchKernelGGL(k_sum_sq_f64, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, cinfo_deg.as<double>(), scalars.as<double>() + 1);
@@ -764,6 +816,11 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     MGX_HIP_TRY(hipMemcpyAsync(exx_a2x, scalars.p, 16, hipMemcpyDeviceToHost, ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
     curr_mod = exx_a2x[0] * constant - exx_a2x[1] * constant * constant;
+    if (louvain_debug() && (iters < 10 || iters % 500 == 0)) {
+      fprintf(stderr, "[louvain] iter=%lld exx=%.12g a2x=%.12g mod=%.12g prev=%.12g\n",
+              (long long)iters, exx_a2x[0], exx_a2x[1], curr_mod, prev_mod);
+      fflush(stderr);
+    }
     if ((curr_mod - prev_mod) < thresh) break;
     prev_mod = curr_mod;
     if (prev_mod < lower) prev_mod = lower;
