@@ -69,7 +69,13 @@ def run_secondary(args):
         print(json.dumps({"error": "no HIP device"}))
         sys.exit(1)
     ctx = nat.init(0)
-    scale = args.scale if args.scale != 26 else 24  # config 3/4 default
+    # config 3/4 defaults: wcc/katz on RMAT-24; louvain on RMAT-22 — at
+    # RMAT-24-weighted the grappolo break rule (modularity gain < 1e-6 per
+    # Jacobi sweep, parallelLouvainMethod.cpp:251) admits thousands of
+    # tiny-gain sweeps on the coarse levels (the reference's own CPU code
+    # crawls the same way there).
+    default_scale = 22 if args.algo == "louvain" else 24
+    scale = args.scale if args.scale != 26 else default_scale
     V = 1 << scale
     E = args.edge_factor * V
     flags = {"wcc": BUILD_SYM_CSR, "katz": BUILD_IN_CSR,
