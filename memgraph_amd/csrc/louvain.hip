@@ -43,11 +43,6 @@ constexpr int kBlock = 256;
 constexpr uint32_t kSmallRowDeg = 256;  // <: wave+LDS table; >=: block+pool
 constexpr int kLdsCap = 512;            // per-wave table entries (>= 2*255)
 
-inline int louvain_env(const char *name) {
-  const char *e = getenv(name);
-  return e ? atoi(e) : 0;
-}
-
 inline bool louvain_debug() {
   static const bool v = [] {
     const char *e = getenv("MGX_LOUVAIN_DEBUG");
