@@ -150,3 +150,26 @@ def test_full_size_properties(nat, ctx):
     assert (r1 >= 0).all()
     assert s1.iterations == 20
     assert s1.sweep_launches == 20
+
+
+def test_scan_order_invariance(nat, ctx, oracle):
+    # The hot-first permutation is internal: results must be identical (to
+    # fp tolerance) to the oracle regardless of input edge order.
+    src, dst = oracle.gen_uniform(3000, 30000, seed=3)
+    perm = np.random.default_rng(0).permutation(len(src))
+    rank_a, _ = gpu_pagerank(nat, ctx, 3000, src, dst, max_iterations=15, eps=0.0)
+    rank_b, _ = gpu_pagerank(nat, ctx, 3000, src[perm], dst[perm], max_iterations=15,
+                             eps=0.0)
+    exp, _ = oracle.pagerank(3000, src, dst, max_iterations=15, eps=0.0)
+    assert np.abs(rank_a - exp).max() <= TOL
+    assert np.abs(rank_b - exp).max() <= TOL
+
+
+def test_stop_epsilon_respected(nat, ctx, oracle):
+    # Default module args: eps=1e-5 must stop at the oracle's iteration.
+    src, dst = oracle.gen_uniform(2000, 20000, seed=4)
+    for eps in (1e-3, 1e-5, 1e-7):
+        rank, stats = gpu_pagerank(nat, ctx, 2000, src, dst, max_iterations=500, eps=eps)
+        exp, iters = oracle.pagerank(2000, src, dst, max_iterations=500, eps=eps)
+        assert stats.iterations == iters, eps
+        assert np.abs(rank - exp).max() <= TOL

@@ -153,3 +153,20 @@ def test_katz_rmat(nat, ctx, oracle):
     assert np.array_equal(np.isfinite(cent), fin)
     if fin.any():
         assert np.abs(cent[fin] - exp[fin]).max() < 1e-9
+
+
+def test_katz_alpha_variants(nat, ctx, oracle):
+    # Bounded-degree graph: several alphas in the convergent regime.
+    rng = np.random.default_rng(41)
+    nv = 600
+    src, dst = [], []
+    for v in range(nv):
+        for k in range(int(rng.integers(1, 6))):
+            src.append(v)
+            dst.append(int((v * 31 + k * 7 + 1) % nv))
+    for alpha in (0.05, 0.1, 0.15):
+        for eps in (1e-2, 1e-3):
+            cent, it = gpu_katz(nat, ctx, nv, src, dst, alpha=alpha, epsilon=eps)
+            exp, it_exp = oracle.katz(nv, src, dst, alpha=alpha, epsilon=eps)
+            assert it == it_exp, (alpha, eps)
+            assert np.abs(cent - exp).max() < 1e-9
