@@ -78,6 +78,16 @@ def test_rmat_unweighted_exact(nat, ctx, oracle):
     assert n == n_exp
 
 
+def test_rmat16_unweighted_exact(nat, ctx, oracle):
+    # Mid-scale exact parity (integer weights => every fp64 sum exact):
+    # exercises multiple coarsening phases and both sweep paths.
+    src, dst = oracle.gen_rmat(16, 8 * (1 << 16), seed=33)
+    comm, n = gpu_louvain(nat, ctx, 1 << 16, src, dst)
+    exp, n_exp = oracle.louvain(1 << 16, src, dst)
+    assert np.array_equal(comm, exp)
+    assert n == n_exp
+
+
 def test_hub_rows_exact(nat, ctx, oracle):
     # Rows above the wave/block degree split (>= 256 neighbours) exercise
     # the global-pool path.
