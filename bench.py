@@ -266,9 +266,7 @@ def main():
     # Roofline for the dominant kernel on THIS rank (rank 0 reports).
     if n_gpus > 1:
         local_rows = shard_range(V, world, rank)[1] - shard_range(V, world, rank)[0]
-        # local edge count ~ E/world on average; use exact local kernel bytes
-        # via E/world approximation is wrong — recompute from the graph:
-        local_edges = E / world  # RMAT dst is near-uniform across ranges at scale 26
+        local_edges = nat.lib.mgx_graph_local_edges(g)  # exact owned edges
         algo_bytes = local_edges * 8 + local_rows * 20
     else:
         algo_bytes = E * 8 + V * 20
@@ -296,6 +294,7 @@ def main():
             "edges": E,
             "damping": args.damping,
             "seed": args.seed,
+            "accumulation": "f64-per-row",
             "parallelism": f"vertex-range shard x{n_gpus}" if n_gpus > 1 else "single",
         },
         "csr_build_ms": csr_build_ms,

@@ -93,6 +93,8 @@ mgx_status mgx_gen_edges_to_host(mgx_context *ctx, int rmat, int scale, int64_t 
 mgx_status mgx_graph_destroy(mgx_context *ctx, mgx_graph *g);
 int64_t mgx_graph_num_vertices(const mgx_graph *g);
 int64_t mgx_graph_num_edges(const mgx_graph *g);
+/* Edges held locally (== num_edges unless vertex-range sharded). */
+int64_t mgx_graph_local_edges(const mgx_graph *g);
 /* Device CSR build time (COO->CSR + degree bins), ms (the "CSR build ms"
  * component of BASELINE.json's metric). */
 double mgx_graph_build_ms(const mgx_graph *g);

@@ -297,6 +297,7 @@ extern "C" mgx_status mgx_graph_destroy(mgx_context *ctx, mgx_graph *g) {
 
 extern "C" int64_t mgx_graph_num_vertices(const mgx_graph *g) { return g->n_vertices; }
 extern "C" int64_t mgx_graph_num_edges(const mgx_graph *g) { return g->n_edges; }
+extern "C" int64_t mgx_graph_local_edges(const mgx_graph *g) { return g->in_edges; }
 extern "C" double mgx_graph_build_ms(const mgx_graph *g) { return g->build_ms; }
 
 extern "C" mgx_status mgx_wcc(mgx_context *ctx, mgx_graph *g, int64_t *out_component,

@@ -47,6 +47,7 @@ class Native:
         self.lib.mgx_graph_build_ms.restype = ctypes.c_double
         self.lib.mgx_graph_num_vertices.restype = ctypes.c_int64
         self.lib.mgx_graph_num_edges.restype = ctypes.c_int64
+        self.lib.mgx_graph_local_edges.restype = ctypes.c_int64
 
     def _check(self, status, what):
         if status != 0:
