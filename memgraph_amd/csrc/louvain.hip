@@ -212,6 +212,9 @@ struct SweepArgs {
   int32_t *pool_keys;
   double *pool_vals;
   const uint64_t *pool_off;  // [n_big+1]
+  uint64_t pool_total;
+  int inkernel_clear;        // experiment flag (MGX_LOUVAIN_INKERNEL_CLEAR)
+  uint32_t *guard_err;       // device flag for region-invariant violations
 };
 
 __device__ inline void wave_lds_fence() {
@@ -330,8 +333,22 @@ __global__ void __launch_bounds__(kBlock) k_sweep_big(SweepArgs A) {
     const uint32_t cap = (uint32_t)(t1 - t0);  // power of two
     int32_t *keys = A.pool_keys + t0;
     double *vals = A.pool_vals + t0;
-    // The pool is cleared before the launch (a per-row in-kernel clear was
-    // tried and implicated in memory faults at RMAT-24; see git history).
+    // Region invariants (diagnosing the in-kernel-clear fault): offsets
+    // monotone, power-of-two capacity, region within the pool.
+    if (A.guard_err && threadIdx.x == 0) {
+      if (t1 <= t0 || t1 > A.pool_total || (cap & (cap - 1)) != 0 ||
+          cap < 2 * (adj2 - adj1))
+        atomicOr(A.guard_err, 1u);
+    }
+    if (A.inkernel_clear) {
+      for (uint32_t sIdx = threadIdx.x; sIdx < cap; sIdx += kBlock) {
+        keys[sIdx] = -1;
+        vals[sIdx] = 0.0;
+      }
+      __syncthreads();
+    }
+    // Otherwise the pool is cleared before the launch (the in-kernel
+    // variant was implicated in memory faults at RMAT-24; see git history).
     if (threadIdx.x == 0) {
       uint32_t h = ((uint32_t)sc * 2654435761u) & (cap - 1);
       keys[h] = sc;
@@ -773,6 +790,19 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   A.pool_keys = pool_keys.as<int32_t>();
   A.pool_vals = pool_vals.as<double>();
   A.pool_off = pool_off.as<uint64_t>();
+  A.pool_total = pool_total;
+  A.inkernel_clear = 0;
+  A.guard_err = nullptr;
+  DevBuf guard;
+  {
+    const char *e = getenv("MGX_LOUVAIN_INKERNEL_CLEAR");
+    if (e && atoi(e)) A.inkernel_clear = 1;
+  }
+  if (louvain_debug()) {
+    MGX_HIP_TRY(guard.alloc(ctx, 4));
+    MGX_HIP_TRY(hipMemsetAsync(guard.p, 0, 4, ctx->stream));
+    A.guard_err = guard.as<uint32_t>();
+  }
 
   double prev_mod = -1.0, curr_mod = -1.0;
   int64_t iters = 0;
@@ -795,10 +825,12 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     }
     MGX_LDBG(ctx, "iter=%lld after-small", (long long)iters);
     if (n_big > 0) {
-      hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for((int64_t)pool_total)),
-                         dim3(kBlock), 0, ctx->stream, (int64_t)pool_total, -1,
-                         pool_keys.as<int32_t>());
-      MGX_HIP_TRY(hipMemsetAsync(pool_vals.p, 0, pool_total * 8, ctx->stream));
+      if (!A.inkernel_clear) {
+        hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for((int64_t)pool_total)),
+                           dim3(kBlock), 0, ctx->stream, (int64_t)pool_total, -1,
+                           pool_keys.as<int32_t>());
+        MGX_HIP_TRY(hipMemsetAsync(pool_vals.p, 0, pool_total * 8, ctx->stream));
+      }
       const int64_t blocks = n_big < 4096 ? n_big : 4096;
       hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
                          A);
@@ -830,6 +862,13 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     if (iters > 100000) break;  // matches runMultiPhaseBasic's totItr cap scale
   }
 
+  if (A.guard_err) {
+    uint32_t ge = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&ge, A.guard_err, 4, hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    fprintf(stderr, "[louvain] GUARD err=%u (nv=%lld)\n", ge, (long long)nv);
+    fflush(stderr);
+  }
   MGX_HIP_TRY(hipMemcpyAsync(C, p_past, nv * 4, hipMemcpyDeviceToDevice, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   *out_mod = curr_mod;
