@@ -9,7 +9,6 @@
 // discovered at its min member — DESIGN.md), so emitted ids are bit-exact.
 
 #include <cstring>
-#include <utility>
 
 #include <rocprim/rocprim.hpp>
 
@@ -41,13 +40,6 @@ struct WccArgs {
   int64_t grid[4];
   int32_t *label;       // in-place (async min-propagation is monotone-safe)
   uint32_t *changed;
-  // Active-set filtering: only rows marked active are swept; a row whose
-  // label drops marks its neighbours (and itself) for the next round.
-  // full != 0 sweeps every row — used as a safety net that guarantees the
-  // final labels are the same global fixpoint as unconditional sweeping.
-  const uint8_t *active;
-  uint8_t *next_active;
-  int full;
 };
 
 template <int LANES>
@@ -64,9 +56,6 @@ __device__ inline bool wcc_rows(const WccArgs &A, int sec, int64_t block_in_sec)
     int32_t row = -1;
     if (ri < nrows) {
       row = rows_list[ri];
-      if (!A.full && !A.active[row]) row = -1;
-    }
-    if (row >= 0) {
       const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
       if constexpr (LANES >= 64) {
         uint32_t s_al = (s + 3u) & ~3u;
@@ -97,32 +86,11 @@ __device__ inline bool wcc_rows(const WccArgs &A, int sec, int64_t block_in_sec)
       __syncthreads();
       if (threadIdx.x == 0) m = min(min(red[0], red[1]), min(red[2], red[3]));
     }
-    bool updated = false;
     if (sub == 0 && row >= 0 && (LANES <= 64 || threadIdx.x == 0)) {
-      // Path-halving candidate: the grandparent label may be lower than
-      // anything in the neighbourhood (accelerates chains like pointer
-      // jumping does).
-      const int32_t l = A.label[row];
-      const int32_t gp = A.label[l];
-      if (gp < m) m = gp;
-      if (m < l) {
+      if (m < A.label[row]) {
         A.label[row] = m;
         changed = true;
-        updated = true;
-        if (A.next_active) A.next_active[row] = 1;
       }
-    }
-    if constexpr (LANES <= 64) {
-      updated = __shfl((int)updated, 0, LANES) != 0;  // group leader broadcast
-    } else {
-      __shared__ int bcast;
-      if (threadIdx.x == 0) bcast = updated ? 1 : 0;
-      __syncthreads();
-      updated = bcast != 0;
-    }
-    if (updated && row >= 0 && A.next_active) {
-      const uint32_t s = A.row_ptr[row], e = A.row_ptr[row + 1];
-      for (uint32_t j = s + sub; j < e; j += LANES) A.next_active[A.col[j]] = 1;
     }
     if constexpr (LANES > 64) __syncthreads();
   }
@@ -148,17 +116,13 @@ __global__ void __launch_bounds__(kBlock) k_wcc_sweep(WccArgs A) {
 
 // Pointer jumping: label[v] <- label[label[v]] until per-pass fixpoint
 // (monotone decreasing; races only skip a shortcut, never break it).
-// A compressed row re-enters the active set; the final full sweep makes
-// any missed neighbour wake-up harmless.
-__global__ void k_wcc_jump(int64_t n, int32_t *label, uint8_t *next_active,
-                           uint32_t *changed) {
+__global__ void k_wcc_jump(int64_t n, int32_t *label, uint32_t *changed) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     const int32_t l = label[i];
     const int32_t ll = label[l];
     if (ll < l) {
       label[i] = ll;
-      if (next_active) next_active[i] = 1;
       atomicOr(changed, 1u);
     }
   }
@@ -195,14 +159,10 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
 
   int32_t *label = nullptr;
   uint32_t *d_changed = nullptr;
-  uint8_t *active = nullptr, *next_active = nullptr;
   MGX_TRY(ctx->alloc_async((void **)&label, V * sizeof(int32_t)));
   MGX_TRY(ctx->alloc_async((void **)&d_changed, sizeof(uint32_t)));
-  MGX_TRY(ctx->alloc_async((void **)&active, V));
-  MGX_TRY(ctx->alloc_async((void **)&next_active, V));
   hipLaunchKernelGGL(k_iota, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                      label);
-  MGX_HIP_TRY(hipMemsetAsync(active, 1, V, ctx->stream));
 
   WccArgs A;
   A.row_ptr = g->sym_row_ptr;
@@ -220,33 +180,16 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
   A.label = label;
   A.changed = d_changed;
 
-  // Active-set rounds until quiescent, then a FULL sweep as the safety net:
-  // the loop only terminates when an unconditional sweep (the original
-  // fixpoint condition) makes no change, so the final labels are identical
-  // to unconditional sweeping — the active set is purely a work filter.
-  bool full_round = false;
-  while (true) {
-    uint32_t h_changed = 0;
+  uint32_t h_changed = 1;
+  while (h_changed) {
     MGX_HIP_TRY(hipMemsetAsync(d_changed, 0, 4, ctx->stream));
-    MGX_HIP_TRY(hipMemsetAsync(next_active, 0, V, ctx->stream));
-    A.active = active;
-    A.next_active = next_active;
-    A.full = full_round ? 1 : 0;
     if (goff > 0)
       hipLaunchKernelGGL(k_wcc_sweep, dim3((uint32_t)goff), dim3(kBlock), 0, ctx->stream, A);
     hipLaunchKernelGGL(k_wcc_jump, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
-                       V, label, next_active, d_changed);
+                       V, label, d_changed);
     MGX_HIP_TRY(hipMemcpyAsync(&h_changed, d_changed, 4, hipMemcpyDeviceToHost,
                                ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-    std::swap(active, next_active);
-    if (h_changed) {
-      full_round = false;  // keep filtering while work remains
-    } else if (!full_round) {
-      full_round = true;   // quiescent under the filter: verify with a full sweep
-    } else {
-      break;               // full sweep made no change: global fixpoint reached
-    }
   }
 
   // Renumber + count.
@@ -285,8 +228,6 @@ mgx_status mgx_wcc_impl(mgx_context *ctx, mgx_graph *g, int64_t *out_component,
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   MGX_TRY(ctx->free_async(label));
   MGX_TRY(ctx->free_async(d_changed));
-  MGX_TRY(ctx->free_async(active));
-  MGX_TRY(ctx->free_async(next_active));
   MGX_TRY(ctx->free_async(flag));
   MGX_TRY(ctx->free_async(scan));
   MGX_TRY(ctx->free_async(d_out));
