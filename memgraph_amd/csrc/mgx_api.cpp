@@ -71,7 +71,14 @@ mgx_status mgx_context::free_async(void *p) {
 
 mgx_status mgx_context::reserve(size_t bytes, void **out) {
   if (bytes > workspace_bytes) {
-    if (workspace) MGX_HIP_TRY(hipFree(workspace));
+    // Drain queued users of the old workspace before freeing it. hipFree
+    // synchronizes the device on ROCm, but that contract is not worth
+    // betting stream-ordered correctness on; growth is rare so the sync
+    // costs nothing.
+    if (workspace) {
+      MGX_HIP_TRY(hipStreamSynchronize(stream));
+      MGX_HIP_TRY(hipFree(workspace));
+    }
     workspace = nullptr;
     workspace_bytes = 0;
     size_t want = bytes + bytes / 2;
