@@ -871,7 +871,11 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   }
   MGX_HIP_TRY(hipMemcpyAsync(C, p_past, nv * 4, hipMemcpyDeviceToDevice, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  *out_mod = curr_mod;
+  // The reference's sweep loop returns prevMod — the (Lower-clamped)
+  // modularity of the ADOPTED assignment, not the last computed currMod
+  // (parallelLouvainMethod.cpp:307). The two differ by < thresh, which
+  // flips the phase-continue decision at loose thresholds.
+  *out_mod = prev_mod;
   *iters_out = iters;
   MGX_HIP_TRY(hipGetLastError());
   return MGX_OK;

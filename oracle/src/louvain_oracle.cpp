@@ -164,7 +164,13 @@ double LouvainLevel(const SymGraph &g, double lower, double thresh, std::vector<
 
   C = past;
   *num_iters = iters;
-  return curr_mod;
+  // parallelLouvainMethod.cpp:307 returns prevMod — the (Lower-clamped)
+  // modularity of the ADOPTED assignment (pastCommAss) — not the last
+  // computed currMod. The difference is < thresh per sweep, but the phase
+  // loop compares it against threshold, so returning currMod flips the
+  // continue/stop decision at loose thresholds (caught by
+  // tests/test_oracle_params.py::test_louvain_threshold_sweep).
+  return prev_mod;
 }
 
 // renumberClustersContiguously (buildNextPhase.cpp:49-78): first-seen order,
