@@ -130,3 +130,21 @@ def test_no_edges(nat, ctx):
     comm, n = gpu_louvain(nat, ctx, 3, [], [])
     assert n == 0
     assert np.all(comm == -1)
+
+
+def test_threshold_sweep_exact(nat, ctx, oracle):
+    # Non-default thresholds (the module's `threshold` argument): the
+    # phase-stop decision depends on the sweep loop returning the
+    # reference's prevMod, not currMod — loose thresholds expose any
+    # mismatch (tests/test_oracle_params.py pins the oracle side).
+    rng = np.random.default_rng(23)
+    for thr in (1e-1, 1e-2, 1e-4):
+        for _ in range(3):
+            nv = int(rng.integers(4, 1200))
+            ne = int(rng.integers(1, 6000))
+            src = rng.integers(0, nv, ne)
+            dst = rng.integers(0, nv, ne)
+            comm, n = gpu_louvain(nat, ctx, nv, src, dst, threshold=thr)
+            exp, n_exp = oracle.louvain(nv, src, dst, threshold=thr)
+            assert np.array_equal(comm, exp), thr
+            assert n == n_exp, thr
