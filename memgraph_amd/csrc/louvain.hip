@@ -215,7 +215,6 @@ struct SweepArgs {
   uint64_t pool_total;
   int inkernel_clear;        // experiment flag (MGX_LOUVAIN_INKERNEL_CLEAR)
   uint32_t *guard_err;       // device flag for region-invariant violations
-  int64_t nv;                // level vertex count (community-id bound)
 };
 
 __device__ inline void wave_lds_fence() {
@@ -306,14 +305,6 @@ __global__ void __launch_bounds__(kBlock) k_sweep_small(SweepArgs A) {
     }
     if (lane == 0) {
       int32_t max_index = (best.gain > 0.0) ? best.cid : sc;
-      // Defensive bound: a corrupt table entry must not become an OOB
-      // cinfo/cupd index (the suspected vector of the flaky storm-regime
-      // fault, DESIGN.md Louvain finding 3) — stay put and flag instead
-      // of faulting. Never fires in a correct execution.
-      if (max_index < 0 || max_index >= A.nv) {
-        if (A.guard_err) atomicAdd(A.guard_err, 1u);
-        max_index = sc;
-      }
       // swap protection (max(), utilityClusteringFunctions.cpp:305-307)
       if (A.cinfo_size[max_index] == 1 && A.cinfo_size[sc] == 1 && max_index > sc)
         max_index = sc;
@@ -420,10 +411,6 @@ __global__ void __launch_bounds__(kBlock) k_sweep_big(SweepArgs A) {
           best = other;
       }
       int32_t max_index = (best.gain > 0.0) ? best.cid : sc;
-      if (max_index < 0 || max_index >= A.nv) {  // defensive bound, see above
-        if (A.guard_err) atomicAdd(A.guard_err, 1u);
-        max_index = sc;
-      }
       if (A.cinfo_size[max_index] == 1 && A.cinfo_size[sc] == 1 && max_index > sc)
         max_index = sc;
       A.target[row] = max_index;
@@ -806,7 +793,6 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   A.pool_total = pool_total;
   A.inkernel_clear = 0;
   A.guard_err = nullptr;
-  A.nv = nv;
   DevBuf guard;
   {
     const char *e = getenv("MGX_LOUVAIN_INKERNEL_CLEAR");
