@@ -126,6 +126,13 @@ mgx_status mgx_pagerank_start(mgx_context *ctx, mgx_graph *g, double damping_fac
                               mgx_pagerank_run **out);
 /* Queue n iterations (async; no stopping rule, eps treated as 0). */
 mgx_status mgx_pagerank_iterate(mgx_pagerank_run *run, int64_t n_iterations);
+/* Iterate with the reference stopping rule (Linf(new-old) <= stop_epsilon,
+ * pagerank.cpp:139-151), up to max_n iterations; *done = iterations run.
+ * Synchronizes once per iteration to read the device Linf. On a
+ * distributed run the Linf is max-allreduced first, so every rank runs the
+ * same iteration count (no collective divergence). */
+mgx_status mgx_pagerank_iterate_eps(mgx_pagerank_run *run, int64_t max_n,
+                                    double stop_epsilon, int64_t *done);
 /* Sweep timing accumulated so far (ms) and launch count. */
 mgx_status mgx_pagerank_timing(mgx_pagerank_run *run, double *sweep_ms, int64_t *launches);
 mgx_status mgx_pagerank_finish(mgx_pagerank_run *run, double *out_rank /* nullable */);

@@ -317,6 +317,14 @@ mgx_status mgx_pagerank_queue_iterations(mgx_pagerank_run *run, int64_t n, bool 
 }
 
 mgx_status mgx_pagerank_read_delta(mgx_pagerank_run *run, float *out) {
+  if (run->dist && mgx_comm_world(run->ctx) > 1) {
+    // The sweep's Linf is over OWNED rows only; ranks must agree on the
+    // stopping decision or the next iteration's ncclAllGather deadlocks.
+    // d_delta holds the f32 bits of a non-negative float (ordered-uint
+    // atomicMax), so reinterpreting as float is exact; in-place max.
+    MGX_TRY(mgx_comm_allreduce_max_f32(run->ctx, (const float *)run->d_delta,
+                                       (float *)run->d_delta));
+  }
   uint32_t bits = 0;
   MGX_HIP_TRY(hipMemcpyAsync(&bits, run->d_delta, 4, hipMemcpyDeviceToHost,
                              run->ctx->stream));
@@ -427,6 +435,22 @@ extern "C" mgx_status mgx_pagerank_start_dist(mgx_context *ctx, mgx_graph *g, do
 
 extern "C" mgx_status mgx_pagerank_iterate(mgx_pagerank_run *run, int64_t n) {
   return mgx_pagerank_queue_iterations(run, n, /*track_delta=*/false);
+}
+
+extern "C" mgx_status mgx_pagerank_iterate_eps(mgx_pagerank_run *run, int64_t max_n,
+                                               double stop_epsilon, int64_t *done) {
+  int64_t i = 0;
+  for (; i < max_n; ++i) {
+    MGX_TRY(mgx_pagerank_queue_iterations(run, 1, /*track_delta=*/true));
+    float delta = 0.f;
+    MGX_TRY(mgx_pagerank_read_delta(run, &delta));  // dist: max-allreduced
+    if (delta <= (float)stop_epsilon) {
+      ++i;
+      break;
+    }
+  }
+  if (done) *done = i;
+  return MGX_OK;
 }
 
 extern "C" mgx_status mgx_pagerank_timing(mgx_pagerank_run *run, double *sweep_ms,

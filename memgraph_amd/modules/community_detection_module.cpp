@@ -12,10 +12,12 @@
 // plus community_detection.get_subgraph(subgraph_nodes: list<node>,
 // subgraph_relationships: list<relationship>, <same optional args>)
 // (community_detection_module.cpp:136-152 / OnSubgraph).
-// coloring / min_graph_shrink / coloring_alg_threshold / num_of_threads are
-// accepted for drop-in compatibility; the GPU path always runs the basic
-// (non-coloring) algorithm, minGraphSize is unused by the reference's basic
+// min_graph_shrink / coloring_alg_threshold / num_of_threads are accepted
+// for drop-in compatibility; minGraphSize is unused by the reference's basic
 // path too (runMultiPhaseBasic ignores it), and thread count is N/A on GPU.
+// coloring=true is REJECTED with an explicit error (the reference's coloring
+// path is a different, coloring-scheduled algorithm — silent substitution of
+// the basic one would be a behavioral divergence).
 
 #include <thread>
 
@@ -60,8 +62,18 @@ void RunLouvain(mgp_graph *graph, mgp_result *result, mgp_memory *memory,
 void OnGraph(mgp_list *args, mgp_graph *graph, mgp_result *result, mgp_memory *memory) {
   try {
     const char *weight_property = ArgString(args, 0);
-    // args 1,2,4,5 accepted, see header comment.
+    // args 2,4,5 accepted, see header comment.
+    const bool coloring = ArgBool(args, 1);
     const double threshold = ArgDouble(args, 3);
+    if (coloring) {
+      // Honest rejection instead of silently running the basic algorithm:
+      // the reference's coloring=true selects grappolo's runMultiPhaseColoring
+      // (louvain.cpp:42-48), a different (coloring-scheduled Gauss-Seidel)
+      // variant whose partitions differ from the basic path's.
+      throw std::runtime_error(
+          "coloring=true is not supported by the GPU backend; "
+          "call community_detection.get(coloring=false)");
+    }
 
     // Louvain numbering: first-seen dense ids (louvain.cpp:86-117) — the
     // observable community numbering depends on it.
@@ -90,7 +102,13 @@ void OnSubgraph(mgp_list *args, mgp_graph *graph, mgp_result *result, mgp_memory
     mgp_list *nodes = ArgList(args, 0);
     mgp_list *relationships = ArgList(args, 1);
     const char *weight_property = ArgString(args, 2);
+    const bool coloring = ArgBool(args, 3);
     const double threshold = ArgDouble(args, 5);
+    if (coloring) {
+      throw std::runtime_error(
+          "coloring=true is not supported by the GPU backend; "
+          "call community_detection.get_subgraph(..., coloring=false)");
+    }
     ScanResult scan = ScanSubgraph(graph, memory, nodes, relationships,
                                    /*read_weights=*/true, weight_property,
                                    kDefaultWeight);

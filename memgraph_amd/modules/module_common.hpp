@@ -290,6 +290,14 @@ inline double ArgDouble(mgp_list *args, size_t i) {
   return out;
 }
 
+inline bool ArgBool(mgp_list *args, size_t i) {
+  mgp_value *v = nullptr;
+  Check(mgp_list_at(args, i, &v), "list_at");
+  int out = 0;
+  Check(mgp_value_get_bool(v, &out), "value_get_bool");
+  return out != 0;
+}
+
 inline const char *ArgString(mgp_list *args, size_t i) {
   mgp_value *v = nullptr;
   Check(mgp_list_at(args, i, &v), "list_at");

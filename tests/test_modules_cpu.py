@@ -124,3 +124,33 @@ def test_module_fails_loudly_without_gpu():
     assert out.returncode == 0, out.stderr
     assert out.stdout.startswith("ERR:"), out.stdout
     assert "no HIP device" in out.stdout
+
+
+COLORING_SCRIPT = r"""
+import sys
+sys.path.insert(0, {mockdir!r})
+from harness import ModuleHost
+h = ModuleHost("community_detection")
+h.load_graph([0, 1], [0], [1])
+h.override_arg(1, True)  # coloring=true
+try:
+    h.call("get")
+    print("NO_ERROR")
+except RuntimeError as e:
+    print("ERR:" + str(e))
+"""
+
+
+def test_community_coloring_rejected():
+    """coloring=true selects a different algorithm in the reference
+    (runMultiPhaseColoring, louvain.cpp:42-48); the GPU backend rejects it
+    explicitly instead of silently running the basic path. The check is
+    before any device work, so it holds on CPU too."""
+    require_built()
+    out = subprocess.run(
+        [sys.executable, "-c",
+         COLORING_SCRIPT.format(mockdir=os.path.join(os.path.dirname(__file__), "mock"))],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    assert out.stdout.startswith("ERR:"), out.stdout
+    assert "coloring" in out.stdout
