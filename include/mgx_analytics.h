@@ -164,6 +164,39 @@ mgx_status mgx_louvain(mgx_context *ctx, mgx_graph *g, double threshold,
 mgx_status mgx_betweenness(mgx_context *ctx, mgx_graph *g, int directed, int normalize,
                            double *out_bc);
 
+/* ---- Online (dynamic) PageRank — replaces pagerank_online_alg
+ *      (query_modules/pagerank_module/algorithm_online/pagerank.cpp:253-318:
+ *      R random walks per node, stop probability epsilon; rank[v] =
+ *      visit_count[v] / sum; updates truncate walks at the changed node and
+ *      regrow the suffix with epsilon/2) ---------------------------------
+ * State is per-process (the reference keeps a global context, pagerank.cpp
+ * :49). Walks live on device: an entry pool of (walk, pos, node-slot)
+ * triples plus per-walk start/liveness; counters are recomputed from live
+ * entries (the reference's walks vector is equally the ground truth, its
+ * counter map a cache). Nodes are keyed by memgraph id via a host slot map
+ * so state survives graph changes between calls.
+ * RNG is counter-based (splitmix64 streams keyed by walk/generation/step)
+ * seeded from `seed`; the reference seeds from std::random_device, so
+ * parity is the statistical bar in DESIGN.md, not bit-equality.
+ * Graph must be built with MGX_BUILD_OUT_CSR; dense_to_mg maps the scan's
+ * dense ids to memgraph ids. out_rank is by dense id. */
+mgx_status mgx_pronline_set(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                            int64_t R, double eps, uint64_t seed, double *out_rank);
+/* consistent=0 <=> some scanned node has no walk state (caller must raise
+ * the reference's inconsistency error, pagerank.cpp:284-288). */
+mgx_status mgx_pronline_get(mgx_context *ctx, const int64_t *dense_to_mg, int64_t V,
+                            double *out_rank, int *consistent);
+mgx_status mgx_pronline_update(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                               const int64_t *created_v, int64_t n_cv,
+                               const int64_t *created_e /* (from,to) mg-id pairs */,
+                               int64_t n_ce, const int64_t *deleted_v, int64_t n_dv,
+                               const int64_t *deleted_e, int64_t n_de, double *out_rank);
+mgx_status mgx_pronline_reset(mgx_context *ctx);
+int mgx_pronline_initialized(void);
+/* Test support: walk-state invariants (counts by live entries). */
+mgx_status mgx_pronline_stats(mgx_context *ctx, int64_t *n_walks, int64_t *n_live_walks,
+                              int64_t *n_live_entries);
+
 /* ---- multi-GPU (RCCL over xGMI) ------------------------- */
 
 #define MGX_UNIQUE_ID_BYTES 128 /* == sizeof(ncclUniqueId) */

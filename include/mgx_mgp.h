@@ -92,10 +92,17 @@ enum mgp_error mgp_value_get_vertex(struct mgp_value *val,
 enum mgp_error mgp_value_get_edge(struct mgp_value *val,
                                   struct mgp_edge **result);                /* :466 */
 
-/* --- lists (mg_procedure.h:554-563) --- */
+/* --- lists (mg_procedure.h:519-563) --- */
+enum mgp_error mgp_list_make_empty(size_t capacity, struct mgp_memory *memory,
+                                   struct mgp_list **result);               /* :519 */
+enum mgp_error mgp_value_make_list(struct mgp_list *val,
+                                   struct mgp_value **result);              /* :256 */
 enum mgp_error mgp_list_size(struct mgp_list *list, size_t *result);        /* :554 */
 enum mgp_error mgp_list_at(struct mgp_list *list, size_t index,
                            struct mgp_value **result);                      /* :563 */
+
+/* enterprise gate used by the online modules (mg_procedure.h:65) */
+int mgp_is_enterprise_valid(void);
 
 /* --- results (mg_procedure.h:716-731) --- */
 enum mgp_error mgp_result_set_error_msg(struct mgp_result *res,
@@ -149,6 +156,8 @@ enum mgp_error mgp_type_node(struct mgp_type **result);                     /* :
 enum mgp_error mgp_type_relationship(struct mgp_type **result);             /* :1756 */
 enum mgp_error mgp_type_list(struct mgp_type *element_type,
                              struct mgp_type **result);                     /* :1765 */
+enum mgp_error mgp_type_nullable(struct mgp_type *type,
+                                 struct mgp_type **result);                 /* :1802 */
 
 /* --- procedure registration (mg_procedure.h:1843-1952) --- */
 typedef void (*mgp_proc_cb)(struct mgp_list *, struct mgp_graph *, struct mgp_result *,

@@ -273,6 +273,14 @@ inline void InsertInt(mgp_result_record *record, const char *field, int64_t v,
   mgp_value_destroy(val);
 }
 
+inline void InsertString(mgp_result_record *record, const char *field, const char *v,
+                         mgp_memory *memory) {
+  mgp_value *val = nullptr;
+  Check(mgp_value_make_string(v, memory, &val), "value_make_string");
+  Check(mgp_result_record_insert(record, field, val), "record_insert(string)");
+  mgp_value_destroy(val);
+}
+
 // Argument readers (procedure-supplied args arrive positionally).
 inline int64_t ArgInt(mgp_list *args, size_t i) {
   mgp_value *v = nullptr;

@@ -13,6 +13,7 @@
 // back result rows.
 
 #include <cstdint>
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <memory>
@@ -166,7 +167,11 @@ mgp_result g_result_handle{&g_result};
 std::vector<std::pair<size_t, Value>> g_arg_overrides;
 
 mgp_type g_t_bool{"bool"}, g_t_string{"string"}, g_t_int{"int"}, g_t_float{"float"},
-    g_t_node{"node"}, g_t_relationship{"relationship"}, g_t_list{"list"};
+    g_t_node{"node"}, g_t_relationship{"relationship"}, g_t_list{"list"},
+    g_t_nullable{"nullable"};
+// Registration-time lists (empty-list defaults for the online update() args)
+// live for the process: ProcArg::default_value copies the raw pointer.
+std::vector<std::unique_ptr<mgp_list>> g_perm_lists;
 }  // namespace
 
 // ---- ABI implementation ------------------------------------------------
@@ -276,6 +281,29 @@ enum mgp_error mgp_value_get_edge(struct mgp_value *val, struct mgp_edge **resul
   if (val->v.kind != Value::kEdge) return MGP_ERROR_LOGIC_ERROR;
   *result = val->v.edge;
   return MGP_ERROR_NO_ERROR;
+}
+
+enum mgp_error mgp_list_make_empty(size_t capacity, struct mgp_memory *,
+                                   struct mgp_list **result) {
+  (void)capacity;
+  g_perm_lists.push_back(std::make_unique<mgp_list>());
+  *result = g_perm_lists.back().get();
+  return MGP_ERROR_NO_ERROR;
+}
+
+enum mgp_error mgp_value_make_list(struct mgp_list *val, struct mgp_value **result) {
+  auto *v = new mgp_value();
+  v->v.kind = Value::kList;
+  v->v.list = val;
+  *result = v;
+  return MGP_ERROR_NO_ERROR;
+}
+
+// Enterprise gate used by the online modules (mg_procedure.h:65). The mock
+// host is "licensed" unless MOCK_ENTERPRISE=0 (for gate tests).
+int mgp_is_enterprise_valid(void) {
+  const char *e = getenv("MOCK_ENTERPRISE");
+  return (e && e[0] == '0') ? 0 : 1;
 }
 
 enum mgp_error mgp_list_size(struct mgp_list *list, size_t *result) {
@@ -446,6 +474,7 @@ enum mgp_error mgp_type_float(struct mgp_type **result) { *result = &g_t_float; 
 enum mgp_error mgp_type_node(struct mgp_type **result) { *result = &g_t_node; return MGP_ERROR_NO_ERROR; }
 enum mgp_error mgp_type_relationship(struct mgp_type **result) { *result = &g_t_relationship; return MGP_ERROR_NO_ERROR; }
 enum mgp_error mgp_type_list(struct mgp_type *, struct mgp_type **result) { *result = &g_t_list; return MGP_ERROR_NO_ERROR; }
+enum mgp_error mgp_type_nullable(struct mgp_type *, struct mgp_type **result) { *result = &g_t_nullable; return MGP_ERROR_NO_ERROR; }
 
 enum mgp_error mgp_module_add_read_procedure(struct mgp_module *module, const char *name,
                                              mgp_proc_cb cb, struct mgp_proc **result) {
@@ -688,6 +717,11 @@ int64_t mock_result_int(int64_t row, const char *field) {
 double mock_result_double(int64_t row, const char *field) {
   auto &f = g_result.rows[row].fields.at(field);
   return f.d;
+}
+
+const char *mock_result_string(int64_t row, const char *field) {
+  auto &f = g_result.rows[row].fields.at(field);
+  return f.s.c_str();
 }
 
 }  // extern "C"

@@ -1,0 +1,249 @@
+"""pagerank_online CPU-side tests: the drop-in .so registers the reference's
+exact procedures (query_modules/pagerank_module/pagerank_online_module.cpp
+:171-260), gates on the enterprise license, resets without a GPU, and the
+seeded oracle restatement satisfies the structural invariants of DESIGN.md's
+statistical-parity bar level 1."""
+import ctypes
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "mock"))
+from harness import MODULES_DIR  # noqa: E402
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+MOCKDIR = os.path.join(os.path.dirname(__file__), "mock")
+
+
+def require_built():
+    if not os.path.exists(os.path.join(MODULES_DIR, "pagerank_online.so")):
+        pytest.skip("pagerank_online.so not built (run make)")
+
+
+REG_SCRIPT = r"""
+import sys, json
+sys.path.insert(0, {mockdir!r})
+from harness import ModuleHost
+h = ModuleHost("pagerank_online")
+print(json.dumps(h.procedures()))
+"""
+
+
+def test_pagerank_online_registration():
+    require_built()
+    out = subprocess.run(
+        [sys.executable, "-c", REG_SCRIPT.format(mockdir=MOCKDIR)],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    import json
+    procs = json.loads(out.stdout)
+    # pagerank_online_module.cpp:176-189
+    assert procs["set"]["args"] == [
+        ["walks_per_node", "int"], ["walk_stop_epsilon", "float"]]
+    assert procs["set"]["results"] == [["node", "node"], ["rank", "float"]]
+    # :197-208
+    assert procs["get"]["args"] == []
+    assert procs["get"]["results"] == [["node", "node"], ["rank", "float"]]
+    # :211-249 (nullable list args)
+    assert procs["update"]["args"] == [
+        ["created_vertices", "nullable"], ["created_edges", "nullable"],
+        ["deleted_vertices", "nullable"], ["deleted_edges", "nullable"]]
+    assert procs["update"]["results"] == [["node", "node"], ["rank", "float"]]
+    # :252-259
+    assert procs["reset"]["args"] == []
+    assert procs["reset"]["results"] == [["message", "string"]]
+
+
+GATE_SCRIPT = r"""
+import sys
+sys.path.insert(0, {mockdir!r})
+from harness import ModuleHost
+h = ModuleHost("pagerank_online")
+h.load_graph([0, 1], [0], [1])
+try:
+    h.call("set")
+    print("NO_ERROR")
+except RuntimeError as e:
+    print("ERR:" + str(e))
+"""
+
+
+def test_enterprise_gate():
+    """pagerank_online_module.cpp:74-77: without a valid enterprise license
+    every procedure errors with the reference's message."""
+    require_built()
+    env = dict(os.environ, MOCK_ENTERPRISE="0")
+    out = subprocess.run(
+        [sys.executable, "-c", GATE_SCRIPT.format(mockdir=MOCKDIR)],
+        capture_output=True, text=True, env=env)
+    assert out.returncode == 0, out.stderr
+    assert "valid enterprise license" in out.stdout
+
+
+RESET_SCRIPT = r"""
+import sys, ctypes
+sys.path.insert(0, {mockdir!r})
+from harness import ModuleHost
+h = ModuleHost("pagerank_online")
+rc = h.mock.mock_call(b"reset")
+h.mock.mock_result_string.restype = ctypes.c_char_p
+print("rc=", rc, "msg=", h.mock.mock_result_string(0, b"message").decode())
+"""
+
+
+def test_reset_without_gpu():
+    """reset only clears walk state; it must succeed on a GPU-less host with
+    the reference's message (pagerank_online_module.cpp:164)."""
+    require_built()
+    out = subprocess.run(
+        [sys.executable, "-c", RESET_SCRIPT.format(mockdir=MOCKDIR)],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr
+    assert "rc= 0" in out.stdout
+    assert "Pagerank context is reset" in out.stdout
+
+
+# ---- seeded oracle restatement: structural invariants --------------------
+
+class Oracle:
+    def __init__(self):
+        self.lib = ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+        self.lib.oracle_pronline_get.restype = ctypes.c_int
+        self.lib.oracle_pronline_initialized.restype = ctypes.c_int
+        I64 = ctypes.c_int64
+        P64 = ctypes.POINTER(ctypes.c_int64)
+        PD = ctypes.POINTER(ctypes.c_double)
+        self.lib.oracle_pronline_set.argtypes = [
+            I64, P64, I64, P64, P64, I64, ctypes.c_double, ctypes.c_uint64, PD]
+        self.lib.oracle_pronline_get.argtypes = [I64, P64, PD]
+        self.lib.oracle_pronline_update.argtypes = [
+            I64, P64, I64, P64, P64, P64, I64, P64, I64, P64, I64, P64, I64, PD]
+
+    @staticmethod
+    def arr(a):
+        x = np.ascontiguousarray(a, dtype=np.int64)
+        return x, x.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))
+
+    def set(self, nodes, src, dst, R=10, eps=0.2, seed=1):
+        nodes, pn = self.arr(nodes)
+        src, ps = self.arr(src)
+        dst, pd = self.arr(dst)
+        rank = np.zeros(len(nodes))
+        self.lib.oracle_pronline_set(
+            len(nodes), pn, len(src), ps, pd, ctypes.c_int64(R),
+            ctypes.c_double(eps), ctypes.c_uint64(seed),
+            rank.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        return rank
+
+    def get(self, nodes):
+        nodes, pn = self.arr(nodes)
+        rank = np.zeros(len(nodes))
+        ok = self.lib.oracle_pronline_get(
+            len(nodes), pn, rank.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        return ok, rank
+
+    def update(self, nodes, src, dst, cv=(), ce=(), dv=(), de=()):
+        nodes, pn = self.arr(nodes)
+        src, ps = self.arr(src)
+        dst, pd = self.arr(dst)
+        cv, pcv = self.arr(cv)
+        ce_flat, pce = self.arr(np.array(ce, dtype=np.int64).reshape(-1))
+        dv, pdv = self.arr(dv)
+        de_flat, pde = self.arr(np.array(de, dtype=np.int64).reshape(-1))
+        rank = np.zeros(len(nodes))
+        self.lib.oracle_pronline_update(
+            len(nodes), pn, len(src), ps, pd, pcv, len(cv), pce, len(ce_flat) // 2,
+            pdv, len(dv), pde, len(de_flat) // 2,
+            rank.ctypes.data_as(ctypes.POINTER(ctypes.c_double)))
+        return rank
+
+    def stats(self):
+        w = ctypes.c_int64(0)
+        v = ctypes.c_int64(0)
+        self.lib.oracle_pronline_stats(ctypes.byref(w), ctypes.byref(v))
+        return w.value, v.value
+
+    def reset(self):
+        self.lib.oracle_pronline_reset()
+
+
+@pytest.fixture
+def orc():
+    o = Oracle()
+    o.reset()
+    yield o
+    o.reset()
+
+
+def test_oracle_set_invariants(orc):
+    # the reference e2e graph (pagerank_online_test/*/input.cyp): triangle
+    # 0->1->2->0 plus 3->3, 3->4, 3->5
+    nodes = [0, 1, 2, 3, 4, 5]
+    src = [0, 1, 2, 3, 3, 3]
+    dst = [1, 2, 0, 3, 4, 5]
+    rank = orc.set(nodes, src, dst, R=10, eps=0.2, seed=7)
+    assert abs(rank.sum() - 1.0) < 1e-12
+    assert (rank > 0).all() and (rank < 1).all()  # the e2e assertion
+    n_walks, visits = orc.stats()
+    assert n_walks == len(nodes) * 10
+    ok, rank2 = orc.get(nodes)
+    assert ok == 1
+    assert np.array_equal(rank, rank2)
+
+
+def test_oracle_inconsistent_get(orc):
+    orc.set([0, 1], [0], [1], seed=3)
+    ok, _ = orc.get([0, 1, 2])  # node 2 has no walk state
+    assert ok == 0
+
+
+def test_oracle_update_new_vertex(orc):
+    orc.set([0, 1], [0], [1], seed=3)
+    # add vertex 2 and edge 1->2 (the reference double_call scenario shape)
+    rank = orc.update([0, 1, 2], [0, 1], [1, 2], cv=[2], ce=[(1, 2)])
+    assert abs(rank.sum() - 1.0) < 1e-12
+    assert (rank > 0).all()
+    n_walks, _ = orc.stats()
+    assert n_walks == 3 * 10
+
+
+def test_oracle_delete_vertex(orc):
+    orc.set([0, 1, 2], [0, 1, 2], [1, 2, 0], seed=5)
+    # detach-delete vertex 2: its edges (2->0 incoming-to-0? outgoing) and 1->2
+    rank = orc.update([0, 1], [0], [1], dv=[2], de=[(1, 2), (2, 0)])
+    assert abs(rank.sum() - 1.0) < 1e-12
+    assert (rank >= 0).all()
+    ok, _ = orc.get([0, 1])
+    assert ok == 1
+
+
+def test_oracle_seed_reproducible(orc):
+    nodes = list(range(8))
+    src = [0, 1, 2, 3, 4, 5, 6, 7]
+    dst = [1, 2, 3, 4, 5, 6, 7, 0]
+    r1 = orc.set(nodes, src, dst, seed=42)
+    orc.reset()
+    r2 = orc.set(nodes, src, dst, seed=42)
+    assert np.array_equal(r1, r2)
+    orc.reset()
+    r3 = orc.set(nodes, src, dst, seed=43)
+    assert not np.array_equal(r1, r3)
+
+
+def test_oracle_statistical_sanity(orc):
+    """Mean online rank over many seeds approximates static PageRank rank
+    ORDER on a well-separated graph (star: hub must rank highest)."""
+    nodes = list(range(6))
+    src = [1, 2, 3, 4, 5]
+    dst = [0, 0, 0, 0, 0]
+    acc = np.zeros(6)
+    S = 30
+    for s in range(S):
+        orc.reset()
+        acc += orc.set(nodes, src, dst, R=10, eps=0.2, seed=100 + s)
+    mean = acc / S
+    assert mean[0] == mean.max()
+    assert mean[0] > 2 * mean[1:].max()
