@@ -199,6 +199,62 @@ class Native:
                                      out.ctypes.data_as(_F64)), "mgx_betweenness")
         return out
 
+    # --- online pagerank (walk state is process-global, like the
+    #     reference's context — algorithm_online/pagerank.cpp:49) ---
+    def pronline_set(self, ctx, g, dense_to_mg, R=10, eps=0.2, seed=1):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        V = len(dense_to_mg)
+        rank = np.zeros(V)
+        self._check(
+            self.lib.mgx_pronline_set(ctx, g, dense_to_mg.ctypes.data_as(_I64),
+                                      ctypes.c_int64(R), ctypes.c_double(eps),
+                                      ctypes.c_uint64(seed),
+                                      rank.ctypes.data_as(_F64)),
+            "mgx_pronline_set")
+        return rank
+
+    def pronline_get(self, ctx, dense_to_mg):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        V = len(dense_to_mg)
+        rank = np.zeros(V)
+        consistent = ctypes.c_int(0)
+        self._check(
+            self.lib.mgx_pronline_get(ctx, dense_to_mg.ctypes.data_as(_I64),
+                                      ctypes.c_int64(V), rank.ctypes.data_as(_F64),
+                                      ctypes.byref(consistent)),
+            "mgx_pronline_get")
+        return rank, consistent.value
+
+    def pronline_update(self, ctx, g, dense_to_mg, cv=(), ce=(), dv=(), de=()):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        V = len(dense_to_mg)
+        cv = np.ascontiguousarray(cv, dtype=np.int64)
+        dv = np.ascontiguousarray(dv, dtype=np.int64)
+        ce = np.ascontiguousarray(np.asarray(ce, dtype=np.int64).reshape(-1))
+        de = np.ascontiguousarray(np.asarray(de, dtype=np.int64).reshape(-1))
+        rank = np.zeros(V)
+        self._check(
+            self.lib.mgx_pronline_update(
+                ctx, g, dense_to_mg.ctypes.data_as(_I64), cv.ctypes.data_as(_I64),
+                ctypes.c_int64(len(cv)), ce.ctypes.data_as(_I64),
+                ctypes.c_int64(len(ce) // 2), dv.ctypes.data_as(_I64),
+                ctypes.c_int64(len(dv)), de.ctypes.data_as(_I64),
+                ctypes.c_int64(len(de) // 2), rank.ctypes.data_as(_F64)),
+            "mgx_pronline_update")
+        return rank
+
+    def pronline_reset(self, ctx):
+        self._check(self.lib.mgx_pronline_reset(ctx), "mgx_pronline_reset")
+
+    def pronline_stats(self, ctx):
+        w = ctypes.c_int64(0)
+        lw = ctypes.c_int64(0)
+        le = ctypes.c_int64(0)
+        self._check(self.lib.mgx_pronline_stats(ctx, ctypes.byref(w), ctypes.byref(lw),
+                                                ctypes.byref(le)),
+                    "mgx_pronline_stats")
+        return w.value, lw.value, le.value
+
     # --- comm ---
     def comm_unique_id(self):
         buf = (ctypes.c_char * UNIQUE_ID_BYTES)()
