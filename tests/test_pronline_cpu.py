@@ -247,3 +247,46 @@ def test_oracle_statistical_sanity(orc):
     mean = acc / S
     assert mean[0] == mean.max()
     assert mean[0] > 2 * mean[1:].max()
+
+
+def test_oracle_distribution_vs_reference():
+    """Pin the seeded oracle's DISTRIBUTION against the reference's own
+    pagerank_online compiled from /root/reference (oracle/_ref/
+    libref_online.so; random_device-seeded inside, so the comparison is
+    Welch on per-node mean ranks over S runs — DESIGN.md bar level 2)."""
+    ref_path = os.path.join(REPO, "oracle", "_ref", "libref_online.so")
+    if not os.path.exists(ref_path):
+        pytest.skip("_ref online lib not built (no /root/reference)")
+    ref = ctypes.CDLL(ref_path)
+    I64 = ctypes.c_int64
+    P64 = ctypes.POINTER(ctypes.c_int64)
+    PD = ctypes.POINTER(ctypes.c_double)
+    ref.ref_pron_set.argtypes = [I64, P64, I64, P64, P64, I64, ctypes.c_double, PD]
+
+    nodes = np.arange(10, dtype=np.int64)
+    src = np.array([1, 2, 3, 4, 5, 5, 6, 7, 8, 9], dtype=np.int64)
+    dst = np.array([0, 0, 0, 0, 0, 6, 7, 8, 9, 5], dtype=np.int64)
+    S = 40
+    V = len(nodes)
+
+    refr = np.zeros((S, V))
+    for s in range(S):
+        ref.ref_pron_reset()
+        out = np.zeros(V)
+        ref.ref_pron_set(V, nodes.ctypes.data_as(P64), len(src),
+                         src.ctypes.data_as(P64), dst.ctypes.data_as(P64),
+                         I64(10), ctypes.c_double(0.2), out.ctypes.data_as(PD))
+        refr[s] = out
+    ref.ref_pron_reset()
+
+    orc = Oracle()
+    ours = np.zeros((S, V))
+    for s in range(S):
+        orc.reset()
+        ours[s] = orc.set(list(nodes), list(src), list(dst), R=10, eps=0.2,
+                          seed=5000 + s)
+    orc.reset()
+
+    se = np.sqrt((refr.var(0, ddof=1) + ours.var(0, ddof=1)) / S) + 1e-15
+    z = np.abs(refr.mean(0) - ours.mean(0)) / se
+    assert (z < 4.0).all(), f"z-scores vs reference: {z}"
