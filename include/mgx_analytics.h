@@ -197,6 +197,27 @@ int mgx_pronline_initialized(void);
 mgx_status mgx_pronline_stats(mgx_context *ctx, int64_t *n_walks, int64_t *n_live_walks,
                               int64_t *n_live_entries);
 
+/* ---- Online (dynamic) Katz centrality — replaces the reference's online
+ *      katz_alg (query_modules/katz_centrality_module/algorithm/katz.cpp:
+ *      SetKatz :356-378, UpdateKatz :380-468, KatzCentralityLoop :211-240).
+ * Deterministic; parity vs the sequential oracle at 1e-9 (f64 sum order).
+ * State is per-process like the reference's context (katz.cpp:105).
+ * Graph needs MGX_BUILD_IN_CSR|MGX_BUILD_OUT_CSR. Created edges arrive as
+ * (from,to) mg-id pairs; skipping by pair multiplicity is value-equivalent
+ * to the reference's skip-by-edge-id (see csrc/katz_online.hip header). */
+mgx_status mgx_konline_set(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                           double alpha, double epsilon, double *out /* [V] */);
+mgx_status mgx_konline_get(mgx_context *ctx, const int64_t *dense_to_mg, int64_t V,
+                           double *out, int *consistent);
+mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                              const int64_t *created_v, int64_t n_cv,
+                              const int64_t *created_e, int64_t n_ce,
+                              const int64_t *deleted_v, int64_t n_dv,
+                              const int64_t *deleted_e, int64_t n_de, double *out);
+mgx_status mgx_konline_reset(mgx_context *ctx);
+int mgx_konline_initialized(void);
+int64_t mgx_konline_iterations(void);
+
 /* ---- multi-GPU (RCCL over xGMI) ------------------------- */
 
 #define MGX_UNIQUE_ID_BYTES 128 /* == sizeof(ncclUniqueId) */

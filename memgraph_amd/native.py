@@ -255,6 +255,55 @@ class Native:
                     "mgx_pronline_stats")
         return w.value, lw.value, le.value
 
+    # --- online katz (state is process-global, like the reference's
+    #     context — query_modules katz.cpp:105) ---
+    def konline_set(self, ctx, g, dense_to_mg, alpha=0.2, eps=1e-2):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        V = len(dense_to_mg)
+        out = np.zeros(V)
+        self._check(
+            self.lib.mgx_konline_set(ctx, g, dense_to_mg.ctypes.data_as(_I64),
+                                     ctypes.c_double(alpha), ctypes.c_double(eps),
+                                     out.ctypes.data_as(_F64)),
+            "mgx_konline_set")
+        return out
+
+    def konline_get(self, ctx, dense_to_mg):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        V = len(dense_to_mg)
+        out = np.zeros(V)
+        consistent = ctypes.c_int(0)
+        self._check(
+            self.lib.mgx_konline_get(ctx, dense_to_mg.ctypes.data_as(_I64),
+                                     ctypes.c_int64(V), out.ctypes.data_as(_F64),
+                                     ctypes.byref(consistent)),
+            "mgx_konline_get")
+        return out, consistent.value
+
+    def konline_update(self, ctx, g, dense_to_mg, cv=(), ce=(), dv=(), de=()):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        cv = np.ascontiguousarray(cv, dtype=np.int64)
+        dv = np.ascontiguousarray(dv, dtype=np.int64)
+        ce = np.ascontiguousarray(np.asarray(ce, dtype=np.int64).reshape(-1))
+        de = np.ascontiguousarray(np.asarray(de, dtype=np.int64).reshape(-1))
+        out = np.zeros(len(dense_to_mg))
+        self._check(
+            self.lib.mgx_konline_update(
+                ctx, g, dense_to_mg.ctypes.data_as(_I64), cv.ctypes.data_as(_I64),
+                ctypes.c_int64(len(cv)), ce.ctypes.data_as(_I64),
+                ctypes.c_int64(len(ce) // 2), dv.ctypes.data_as(_I64),
+                ctypes.c_int64(len(dv)), de.ctypes.data_as(_I64),
+                ctypes.c_int64(len(de) // 2), out.ctypes.data_as(_F64)),
+            "mgx_konline_update")
+        return out
+
+    def konline_reset(self, ctx):
+        self._check(self.lib.mgx_konline_reset(ctx), "mgx_konline_reset")
+
+    def konline_iterations(self):
+        self.lib.mgx_konline_iterations.restype = ctypes.c_int64
+        return self.lib.mgx_konline_iterations()
+
     # --- comm ---
     def comm_unique_id(self):
         buf = (ctypes.c_char * UNIQUE_ID_BYTES)()

@@ -1,0 +1,213 @@
+"""GPU tests for katz_centrality_online.
+
+The online Katz algorithm is deterministic, so parity is the ordinary
+oracle bar: values within 1e-9 of the sequential restatement (which is
+itself pinned at 1e-12 against the reference core compiled from
+/root/reference — tests/test_konline_cpu.py)."""
+import os
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.dirname(__file__))
+
+from memgraph_amd.native import BUILD_IN_CSR, BUILD_OUT_CSR, Native  # noqa: E402
+from test_konline_cpu import KOracle  # noqa: E402
+import ctypes  # noqa: E402
+
+pytestmark = pytest.mark.gpu
+
+FLAGS = BUILD_IN_CSR | BUILD_OUT_CSR
+
+
+@pytest.fixture(scope="module")
+def nat():
+    n = Native()
+    if n.device_count() == 0:
+        pytest.skip("no HIP device")
+    return n
+
+
+@pytest.fixture(scope="module")
+def ctx(nat):
+    c = nat.init(0)
+    yield c
+    nat.destroy(c)
+
+
+@pytest.fixture
+def orc():
+    o = KOracle(ctypes.CDLL(os.path.join(REPO, "oracle", "liboracle.so")),
+                "oracle_konline_")
+    o.reset()
+    yield o
+    o.reset()
+
+
+NODES = [0, 1, 2, 3, 4, 5]
+SRC = [0, 1, 2, 3, 3, 3]
+DST = [1, 2, 0, 3, 4, 5]
+
+
+def make_graph(nat, ctx, nodes, src, dst):
+    remap = {m: i for i, m in enumerate(nodes)}
+    s = [remap[x] for x in src]
+    d = [remap[x] for x in dst]
+    if len(nodes) == 0:
+        return None
+    return nat.graph_from_coo(ctx, s, d, len(nodes), flags=FLAGS)
+
+
+def assert_close(a, b, tol=1e-9):
+    assert np.abs(a - b).max() <= tol, f"max diff {np.abs(a - b).max()}"
+
+
+def test_konline_set_parity(nat, ctx, orc):
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, NODES, SRC, DST)
+    got = nat.konline_set(ctx, g, NODES)
+    exp = orc.set(NODES, SRC, DST)
+    assert_close(got, exp)
+    got2, consistent = nat.konline_get(ctx, NODES)
+    assert consistent == 1
+    assert np.array_equal(got, got2)
+    nat.graph_destroy(ctx, g)
+
+
+def test_konline_set_parity_random(nat, ctx, orc):
+    rng = np.random.RandomState(11)
+    V, E = 60, 240
+    src = list(rng.randint(0, V, E))
+    dst = list(rng.randint(0, V, E))
+    nodes = list(range(V))
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, nodes, src, dst)
+    got = nat.konline_set(ctx, g, nodes, alpha=0.1, eps=1e-3)
+    exp = orc.set(nodes, src, dst, alpha=0.1, eps=1e-3)
+    assert_close(got, exp)
+    nat.graph_destroy(ctx, g)
+
+
+def test_konline_get_inconsistent(nat, ctx):
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, [0, 1], [0], [1])
+    nat.konline_set(ctx, g, [0, 1])
+    _, consistent = nat.konline_get(ctx, [0, 1, 2])
+    assert consistent == 0
+    # and the other direction (state node no longer in graph)
+    _, consistent = nat.konline_get(ctx, [0])
+    assert consistent == 0
+    nat.graph_destroy(ctx, g)
+
+
+def test_konline_update_add_edge(nat, ctx, orc):
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, NODES, SRC, DST)
+    nat.konline_set(ctx, g, NODES)
+    orc.set(NODES, SRC, DST)
+    nat.graph_destroy(ctx, g)
+
+    src2 = SRC + [4]
+    dst2 = DST + [5]
+    g2 = make_graph(nat, ctx, NODES, src2, dst2)
+    got = nat.konline_update(ctx, g2, NODES, ce=[(4, 5)])
+    exp = orc.update(NODES, src2, dst2, ce=[(4, 5)], ce_idx=[len(SRC)])
+    assert_close(got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_konline_update_add_vertex(nat, ctx, orc):
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, NODES, SRC, DST)
+    nat.konline_set(ctx, g, NODES)
+    orc.set(NODES, SRC, DST)
+    nat.graph_destroy(ctx, g)
+
+    nodes2 = NODES + [6]
+    src2 = SRC + [4]
+    dst2 = DST + [6]
+    g2 = make_graph(nat, ctx, nodes2, src2, dst2)
+    got = nat.konline_update(ctx, g2, nodes2, cv=[6], ce=[(4, 6)])
+    exp = orc.update(nodes2, src2, dst2, cv=[6], ce=[(4, 6)], ce_idx=[len(SRC)])
+    assert_close(got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_konline_update_delete_edge(nat, ctx, orc):
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, NODES, SRC, DST)
+    nat.konline_set(ctx, g, NODES)
+    orc.set(NODES, SRC, DST)
+    nat.graph_destroy(ctx, g)
+
+    keep = [i for i in range(len(SRC)) if not (SRC[i] == 3 and DST[i] == 4)]
+    src2 = [SRC[i] for i in keep]
+    dst2 = [DST[i] for i in keep]
+    g2 = make_graph(nat, ctx, NODES, src2, dst2)
+    got = nat.konline_update(ctx, g2, NODES, de=[(3, 4)])
+    exp = orc.update(NODES, src2, dst2, de=[(3, 4)])
+    assert_close(got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_konline_update_delete_vertex(nat, ctx, orc):
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, NODES, SRC, DST)
+    nat.konline_set(ctx, g, NODES)
+    orc.set(NODES, SRC, DST)
+    nat.graph_destroy(ctx, g)
+
+    nodes2 = [0, 1, 2, 3, 4]
+    keep = [i for i in range(len(SRC)) if DST[i] != 5 and SRC[i] != 5]
+    src2 = [SRC[i] for i in keep]
+    dst2 = [DST[i] for i in keep]
+    g2 = make_graph(nat, ctx, nodes2, src2, dst2)
+    got = nat.konline_update(ctx, g2, nodes2, dv=[5], de=[(3, 5)])
+    exp = orc.update(nodes2, src2, dst2, dv=[5], de=[(3, 5)])
+    assert_close(got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_konline_update_multi_edge(nat, ctx, orc):
+    nat.konline_reset(ctx)
+    src0 = SRC + [0]
+    dst0 = DST + [1]
+    g = make_graph(nat, ctx, NODES, src0, dst0)
+    nat.konline_set(ctx, g, NODES)
+    orc.set(NODES, src0, dst0)
+    nat.graph_destroy(ctx, g)
+
+    src2 = src0 + [0]
+    dst2 = dst0 + [1]
+    g2 = make_graph(nat, ctx, NODES, src2, dst2)
+    got = nat.konline_update(ctx, g2, NODES, ce=[(0, 1)])
+    exp = orc.update(NODES, src2, dst2, ce=[(0, 1)], ce_idx=[len(src0)])
+    assert_close(got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_konline_sequential_updates(nat, ctx, orc):
+    rng = np.random.RandomState(17)
+    V, E = 24, 72
+    src = list(rng.randint(0, V, E))
+    dst = list(rng.randint(0, V, E))
+    nodes = list(range(V))
+    nat.konline_reset(ctx)
+    g = make_graph(nat, ctx, nodes, src, dst)
+    nat.konline_set(ctx, g, nodes, alpha=0.15, eps=1e-3)
+    orc.set(nodes, src, dst, alpha=0.15, eps=1e-3)
+    nat.graph_destroy(ctx, g)
+    for step in range(3):
+        s, d = int(rng.randint(0, V)), int(rng.randint(0, V))
+        src2 = src + [s]
+        dst2 = dst + [d]
+        g2 = make_graph(nat, ctx, nodes, src2, dst2)
+        got = nat.konline_update(ctx, g2, nodes, ce=[(s, d)])
+        exp = orc.update(nodes, src2, dst2, ce=[(s, d)], ce_idx=[len(src)])
+        assert_close(got, exp)
+        nat.graph_destroy(ctx, g2)
+        src, dst = src2, dst2
+    nat.konline_reset(ctx)
