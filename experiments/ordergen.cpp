@@ -82,6 +82,33 @@ int main(int argc, char **argv) {
     for (size_t k = 0; k < tail.size(); ++k) perm[tail[k]] = (int32_t)(K + k);
   }
 
+  // hot4_xcd3: like hot4 but the tail orders by (top-3 original-id bits,
+  // degree desc) — the RMAT bit-prefix correlation groups sources gathered
+  // by same-prefix destinations — and bin row lists are written with
+  // POSITION-AWARE XCD filling: rows whose original id shares a top-3
+  // prefix land on list positions whose covering block maps to the same
+  // XCD (b % 8), so each XCD's 4 MB L2 caches its own source cluster.
+  const bool xcd_mode = !strcmp(mode, "hot4_xcd3");
+  if (xcd_mode) {
+    const int64_t K = 1 << 20;
+    std::vector<int32_t> permdd(V);
+    for (int64_t k = 0; k < V; ++k) permdd[order[k]] = (int32_t)k;
+    std::vector<uint8_t> is_hot(V, 0);
+    for (int64_t k = 0; k < K; ++k) is_hot[order[k]] = 1;
+    std::vector<int32_t> tail;
+    tail.reserve(V - K);
+    for (int64_t v = 0; v < V; ++v)
+      if (!is_hot[v]) tail.push_back((int32_t)v);
+    const int shift = scale - 3;
+    __gnu_parallel::stable_sort(tail.begin(), tail.end(), [&](int32_t a, int32_t b) {
+      const int ca = a >> shift, cb = b >> shift;
+      if (ca != cb) return ca < cb;
+      return outdeg[a] > outdeg[b];
+    });
+    for (int64_t k = 0; k < K; ++k) perm[order[k]] = (int32_t)k;
+    for (size_t k = 0; k < tail.size(); ++k) perm[tail[k]] = (int32_t)(K + k);
+  }
+
   // permuted sorted CSR
   std::vector<uint64_t> keys(E);
 #pragma omp parallel for
@@ -99,6 +126,51 @@ int main(int argc, char **argv) {
   std::vector<int32_t> col(E);
 #pragma omp parallel for
   for (int64_t i = 0; i < E; ++i) col[i] = (int32_t)(uint32_t)keys[i];
+
+  if (xcd_mode) {
+    // orig id per permuted row
+    std::vector<int32_t> orig(V);
+    for (int64_t v = 0; v < V; ++v) orig[perm[v]] = (int32_t)v;
+    const int shift = scale - 3;
+    const long rpb[4] = {64, 16, 4, 1};
+    const long capb[4] = {2048, 2048, 2048, 8192};
+    for (int b = 0; b < 4; ++b) {
+      std::vector<int32_t> binrows;
+      for (int64_t r = 0; r < V; ++r) {
+        const uint32_t d = row_ptr[r + 1] - row_ptr[r];
+        const int k = d < 8 ? 0 : d < 64 ? 1 : d < 1024 ? 2 : 3;
+        if (k == b) binrows.push_back((int32_t)r);
+      }
+      const long cnt = (long)binrows.size();
+      long need = (cnt + rpb[b] - 1) / rpb[b];
+      const long grid = cnt ? (need < capb[b] ? need : capb[b]) : 0;
+      // cluster queues (rows in ascending permuted id within cluster)
+      std::vector<std::vector<int32_t>> q(8);
+      for (auto r : binrows) q[orig[r] >> shift].push_back(r);
+      std::vector<size_t> head(8, 0);
+      std::vector<int32_t> filled(cnt);
+      const long seg = grid * rpb[b];
+      for (long p = 0; p < cnt; ++p) {
+        const int x = (int)(((p % (seg > 0 ? seg : 1)) / rpb[b]) % 8);
+        int pick = -1;
+        if (head[x] < q[x].size()) {
+          pick = x;
+        } else {  // spill: take from the fullest remaining cluster
+          size_t best = 0;
+          for (int c = 0; c < 8; ++c) {
+            const size_t rem = q[c].size() - head[c];
+            if (rem > best) { best = rem; pick = c; }
+          }
+        }
+        filled[p] = q[pick][head[pick]++];
+      }
+      char bp[512];
+      snprintf(bp, sizeof bp, "%s.rows%d.bin", out, b);
+      FILE *bf = fopen(bp, "wb");
+      fwrite(filled.data(), 4, cnt, bf);
+      fclose(bf);
+    }
+  }
 
   char path[512];
   snprintf(path, sizeof path, "%s.row_ptr.bin", out);

@@ -75,9 +75,13 @@ typedef struct {
 } Block;
 
 int main(int argc, char **argv) {
-  if (argc < 4) { fprintf(stderr, "usage: sim row_ptr.bin col.bin V [--no-colstream]\n"); return 2; }
+  if (argc < 4) { fprintf(stderr, "usage: sim row_ptr.bin col.bin V [--no-colstream] [binprefix]\n"); return 2; }
   int colstream = 1;
-  if (argc > 4 && !strcmp(argv[4], "--no-colstream")) colstream = 0;
+  const char *binprefix = NULL;
+  for (int a = 4; a < argc; ++a) {
+    if (!strcmp(argv[a], "--no-colstream")) colstream = 0;
+    else binprefix = argv[a];
+  }
   long V = atol(argv[3]);
   FILE *f = fopen(argv[1], "rb");
   uint32_t *row_ptr = malloc((V + 1) * 4);
@@ -98,11 +102,23 @@ int main(int argc, char **argv) {
   }
   int32_t *rows[4];
   for (int b = 0; b < 4; ++b) rows[b] = malloc((cnt[b] ? cnt[b] : 1) * 4);
+  if (binprefix) {
+    // custom bin row lists (position-aware XCD clustering experiments)
+    for (int b = 0; b < 4; ++b) {
+      char p[512];
+      snprintf(p, sizeof p, "%s.rows%d.bin", binprefix, b);
+      FILE *bf = fopen(p, "rb");
+      if (!bf) { fprintf(stderr, "missing %s\n", p); return 4; }
+      if (cnt[b] > 0 && fread(rows[b], 4, cnt[b], bf) != (size_t)cnt[b]) return 5;
+      fclose(bf);
+    }
+  } else {
   long fill[4] = {0, 0, 0, 0};
   for (long r = 0; r < V; ++r) {
     uint32_t d = row_ptr[r + 1] - row_ptr[r];
     int k = d < 8 ? 0 : d < 64 ? 1 : d < 1024 ? 2 : 3;
     rows[k][fill[k]++] = (int32_t)r;
+  }
   }
   const long rpb[4] = {64, 16, 4, 1};
   const long cap[4] = {2048, 2048, 2048, 8192};

@@ -57,6 +57,10 @@ mgx_status mgx_sync(mgx_context *ctx);
 #define MGX_BUILD_SYM_CSR 2u
 #define MGX_BUILD_WEIGHTED 4u /* with SYM_CSR: carry fp32 edge weights */
 #define MGX_BUILD_OUT_CSR 8u /* row = source, cols = destinations (Brandes) */
+#define MGX_BUILD_NO_PERM 16u /* identity vertex layout: skip the hot-first \
+                               * renumbering (online/dynamic state keyed by \
+                               * scan ids needs stable row ids across calls). \
+                               * With IN_CSR|WEIGHTED also builds in_w. */
 
 /* Upload a host COO and build the requested CSRs on device.
  * Replaces the layouts built by pagerank_alg::PageRankGraph
@@ -217,6 +221,29 @@ mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g, const int64_t *den
 mgx_status mgx_konline_reset(mgx_context *ctx);
 int mgx_konline_initialized(void);
 int64_t mgx_konline_iterations(void);
+
+/* ---- LabelRankT online community detection — replaces LabelRankT
+ *      (query_modules/community_detection_module/algorithm_online/
+ *      community_detection.cpp; SetLabels :311-328, UpdateLabels :330-351,
+ *      GetLabels :305-309). Deterministic; parity = identical labels vs the
+ *      sequential oracle (pinned against the compiled reference core).
+ * Graph: undirected -> MGX_BUILD_SYM_CSR (|WEIGHTED); directed ->
+ * MGX_BUILD_IN_CSR|MGX_BUILD_NO_PERM (|WEIGHTED). out_label is by dense id,
+ * renumbered 1..k as AllLabels does (-1 = unlabeled). */
+mgx_status mgx_lrt_set(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                       int directed, int weighted, double similarity_threshold,
+                       double exponent, double min_value, double w_selfloop,
+                       int64_t max_iterations, int64_t max_updates, int64_t *out_label);
+/* ran_set=1 <=> state was uncalculated and a full (non-persisted) compute
+ * ran, as GetLabels does. */
+mgx_status mgx_lrt_get(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                       int64_t *out_label, int *ran_set);
+mgx_status mgx_lrt_update(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
+                          const int64_t *mod_v, int64_t n_mv, const int64_t *mod_e,
+                          int64_t n_me, const int64_t *del_v, int64_t n_dv,
+                          const int64_t *del_e, int64_t n_de, int64_t *out_label);
+mgx_status mgx_lrt_reset(mgx_context *ctx);
+int mgx_lrt_initialized(void);
 
 /* ---- multi-GPU (RCCL over xGMI) ------------------------- */
 

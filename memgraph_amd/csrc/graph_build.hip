@@ -201,6 +201,44 @@ mgx_status build_sorted_cols(mgx_context *ctx, const int32_t *d_src, const int32
   return MGX_OK;
 }
 
+// Weighted variant of build_sorted_cols (identity layout only): carries the
+// fp32 edge weight through the (dst<<32|src) sort so in_w[j] matches
+// in_col[j]. Keys sort on the full 64 bits so weight order is deterministic
+// (ties between parallel edges keep ascending weight-payload order — the
+// weights of parallel edges are interchangeable for every consumer anyway).
+mgx_status build_sorted_cols_w(mgx_context *ctx, const int32_t *d_src,
+                               const int32_t *d_dst, const float *d_w, int64_t n_edges,
+                               int64_t key_rows, int32_t *col, float *out_w) {
+  if (n_edges == 0) return MGX_OK;
+  uint64_t *keys = nullptr, *keys_out = nullptr;
+  float *vals_out = nullptr;
+  MGX_TRY(ctx->alloc_async((void **)&keys, n_edges * sizeof(uint64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&keys_out, n_edges * sizeof(uint64_t)));
+  MGX_TRY(ctx->alloc_async((void **)&vals_out, n_edges * sizeof(float)));
+  hipLaunchKernelGGL(k_pack_pairs, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                     n_edges, d_src, d_dst, (const int32_t *)nullptr, keys);
+  int end_bit = 33;
+  while ((1ll << (end_bit - 32)) < key_rows + 1) ++end_bit;
+  size_t tmp_bytes = 0;
+  auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys, keys_out,
+                                       (const float *)d_w, vals_out, n_edges, 0, end_bit,
+                                       ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  void *tmp = nullptr;
+  MGX_TRY(ctx->reserve(tmp_bytes, &tmp));
+  err = rocprim::radix_sort_pairs(tmp, tmp_bytes, keys, keys_out, (const float *)d_w,
+                                  vals_out, n_edges, 0, end_bit, ctx->stream);
+  if (err != hipSuccess) return MGX_ERR_HIP;
+  hipLaunchKernelGGL(k_unpack_cols, dim3(grid_for(n_edges)), dim3(kBlock), 0, ctx->stream,
+                     n_edges, keys_out, col);
+  MGX_HIP_TRY(hipMemcpyAsync(out_w, vals_out, n_edges * sizeof(float),
+                             hipMemcpyDeviceToDevice, ctx->stream));
+  MGX_TRY(ctx->free_async(keys));
+  MGX_TRY(ctx->free_async(keys_out));
+  MGX_TRY(ctx->free_async(vals_out));
+  return MGX_OK;
+}
+
 __global__ void k_pack_sym(int64_t n_edges, const int32_t *src, const int32_t *dst,
                            const float *w, uint64_t *keys, float *vals) {
   // Each input edge twice: (s->d) and (d->s), the GetGrappoloSuitableGraph
@@ -500,7 +538,23 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
   hipLaunchKernelGGL(k_inv_outdeg, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                      g->out_degree, g->inv_outdeg);
 
-  if (flags & MGX_BUILD_IN_CSR) {
+  if ((flags & MGX_BUILD_IN_CSR) && (flags & MGX_BUILD_NO_PERM)) {
+    // Identity layout (online/dynamic paths: row ids must equal scan ids).
+    MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
+    hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_dst,
+                       counts);
+    MGX_HIP_TRY(hipMalloc(&g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_TRY(scan_counts(ctx, counts, V, g->in_row_ptr));
+    MGX_HIP_TRY(hipMalloc(&g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
+    if ((flags & MGX_BUILD_WEIGHTED) && d_w) {
+      MGX_HIP_TRY(hipMalloc(&g->in_w, (E > 0 ? E : 1) * sizeof(float)));
+      MGX_TRY(build_sorted_cols_w(ctx, d_src, d_dst, d_w, E, V, g->in_col, g->in_w));
+    } else {
+      MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, nullptr, E, V, false, 0, 0, g->in_col,
+                                E));
+    }
+    MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, V, &g->bins_in));
+  } else if (flags & MGX_BUILD_IN_CSR) {
     // Hot-first vertex permutation: renumber by descending out-degree
     // (stable, ties by original id) so the most-gathered contrib entries
     // pack into the lowest addresses (L2/L3-resident under power-law skew).

@@ -287,6 +287,7 @@ extern "C" mgx_status mgx_graph_destroy(mgx_context *ctx, mgx_graph *g) {
   if (g->inv_outdeg) (void)hipFree(g->inv_outdeg);
   if (g->out_row_ptr) (void)hipFree(g->out_row_ptr);
   if (g->out_col) (void)hipFree(g->out_col);
+  if (g->in_w) (void)hipFree(g->in_w);
   if (g->bins_out.rows) (void)hipFree(g->bins_out.rows);
   if (g->sym_row_ptr) (void)hipFree(g->sym_row_ptr);
   if (g->sym_col) (void)hipFree(g->sym_col);

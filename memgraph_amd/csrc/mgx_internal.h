@@ -93,6 +93,10 @@ struct mgx_graph {
   uint32_t *out_row_ptr = nullptr;
   int32_t *out_col = nullptr;
 
+  // in-CSR edge weights, built only with MGX_BUILD_NO_PERM|WEIGHTED
+  // (directed LabelRankT).
+  float *in_w = nullptr;
+
   // symmetric CSR (WCC/Louvain): each input edge twice.
   uint32_t *sym_row_ptr = nullptr;
   int32_t *sym_col = nullptr;

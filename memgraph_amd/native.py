@@ -15,6 +15,7 @@ BUILD_IN_CSR = 1
 BUILD_SYM_CSR = 2
 BUILD_WEIGHTED = 4
 BUILD_OUT_CSR = 8
+BUILD_NO_PERM = 16
 
 UNIQUE_ID_BYTES = 128
 
@@ -303,6 +304,56 @@ class Native:
     def konline_iterations(self):
         self.lib.mgx_konline_iterations.restype = ctypes.c_int64
         return self.lib.mgx_konline_iterations()
+
+    # --- LabelRankT online community detection (process-global state) ---
+    def lrt_set(self, ctx, g, dense_to_mg, directed=False, weighted=False,
+                similarity_threshold=0.7, exponent=4.0, min_value=0.1,
+                w_selfloop=1.0, max_iterations=100, max_updates=5):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        out = np.full(len(dense_to_mg), -1, dtype=np.int64)
+        self._check(
+            self.lib.mgx_lrt_set(ctx, g, dense_to_mg.ctypes.data_as(_I64),
+                                 ctypes.c_int(1 if directed else 0),
+                                 ctypes.c_int(1 if weighted else 0),
+                                 ctypes.c_double(similarity_threshold),
+                                 ctypes.c_double(exponent),
+                                 ctypes.c_double(min_value),
+                                 ctypes.c_double(w_selfloop),
+                                 ctypes.c_int64(max_iterations),
+                                 ctypes.c_int64(max_updates),
+                                 out.ctypes.data_as(_I64)),
+            "mgx_lrt_set")
+        return out
+
+    def lrt_get(self, ctx, g, dense_to_mg):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        out = np.full(len(dense_to_mg), -1, dtype=np.int64)
+        ran = ctypes.c_int(0)
+        self._check(
+            self.lib.mgx_lrt_get(ctx, g, dense_to_mg.ctypes.data_as(_I64),
+                                 out.ctypes.data_as(_I64), ctypes.byref(ran)),
+            "mgx_lrt_get")
+        return out, ran.value
+
+    def lrt_update(self, ctx, g, dense_to_mg, mv=(), me=(), dv=(), de=()):
+        dense_to_mg = np.ascontiguousarray(dense_to_mg, dtype=np.int64)
+        mv = np.ascontiguousarray(mv, dtype=np.int64)
+        dv = np.ascontiguousarray(dv, dtype=np.int64)
+        me = np.ascontiguousarray(np.asarray(me, dtype=np.int64).reshape(-1))
+        de = np.ascontiguousarray(np.asarray(de, dtype=np.int64).reshape(-1))
+        out = np.full(len(dense_to_mg), -1, dtype=np.int64)
+        self._check(
+            self.lib.mgx_lrt_update(
+                ctx, g, dense_to_mg.ctypes.data_as(_I64), mv.ctypes.data_as(_I64),
+                ctypes.c_int64(len(mv)), me.ctypes.data_as(_I64),
+                ctypes.c_int64(len(me) // 2), dv.ctypes.data_as(_I64),
+                ctypes.c_int64(len(dv)), de.ctypes.data_as(_I64),
+                ctypes.c_int64(len(de) // 2), out.ctypes.data_as(_I64)),
+            "mgx_lrt_update")
+        return out
+
+    def lrt_reset(self, ctx):
+        self._check(self.lib.mgx_lrt_reset(ctx), "mgx_lrt_reset")
 
     # --- comm ---
     def comm_unique_id(self):

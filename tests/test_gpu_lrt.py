@@ -1,0 +1,237 @@
+"""GPU tests for community_detection_online / LabelRankT.
+
+LabelRankT is deterministic (no RNG), so the parity bar is EXACT label
+equality against the reference's own LabelRankT core compiled from
+/root/reference into oracle/_ref/libref_online.so — the binary travels to
+the GPU box (oracle/_ref is git-ignored but not gpurun-ignored), and for
+this row the compiled reference IS the oracle."""
+import ctypes
+import os
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+from memgraph_amd.native import (BUILD_IN_CSR, BUILD_NO_PERM, BUILD_SYM_CSR,
+                                 BUILD_WEIGHTED, Native)  # noqa: E402
+
+pytestmark = pytest.mark.gpu
+
+I64 = ctypes.c_int64
+P64 = ctypes.POINTER(ctypes.c_int64)
+PD = ctypes.POINTER(ctypes.c_double)
+
+
+class RefLrt:
+    def __init__(self):
+        path = os.path.join(REPO, "oracle", "_ref", "libref_online.so")
+        if not os.path.exists(path):
+            pytest.skip("_ref online lib not built")
+        self.lib = ctypes.CDLL(path)
+        self.lib.ref_lrt_set.argtypes = [
+            I64, P64, I64, P64, P64, PD, ctypes.c_int32, ctypes.c_int32,
+            ctypes.c_double, ctypes.c_double, ctypes.c_double, ctypes.c_double,
+            I64, I64, P64]
+        self.lib.ref_lrt_update.argtypes = [
+            I64, P64, I64, P64, P64, PD, ctypes.c_int32, P64, I64, P64, I64,
+            P64, I64, P64, I64, P64]
+
+    @staticmethod
+    def arr(a, dt=np.int64):
+        x = np.ascontiguousarray(a, dtype=dt)
+        return x
+
+    def set(self, nodes, src, dst, weights=None, directed=False, weighted=False,
+            sim_th=0.7, exponent=4.0, min_value=0.1, w_selfloop=1.0,
+            max_iterations=100, max_updates=5):
+        nodes = self.arr(nodes)
+        src = self.arr(src)
+        dst = self.arr(dst)
+        w = self.arr(weights if weights is not None else [], np.float64)
+        out = np.full(len(nodes), -1, dtype=np.int64)
+        self.lib.ref_lrt_set(
+            len(nodes), nodes.ctypes.data_as(P64), len(src),
+            src.ctypes.data_as(P64), dst.ctypes.data_as(P64),
+            w.ctypes.data_as(PD) if weights is not None else None,
+            1 if directed else 0, 1 if weighted else 0, sim_th, exponent,
+            min_value, w_selfloop, I64(max_iterations), I64(max_updates),
+            out.ctypes.data_as(P64))
+        return out
+
+    def update(self, nodes, src, dst, weights=None, directed=False, mv=(), me=(),
+               dv=(), de=()):
+        nodes = self.arr(nodes)
+        src = self.arr(src)
+        dst = self.arr(dst)
+        w = self.arr(weights if weights is not None else [], np.float64)
+        mv = self.arr(mv)
+        dv = self.arr(dv)
+        me = self.arr(np.asarray(me, dtype=np.int64).reshape(-1))
+        de = self.arr(np.asarray(de, dtype=np.int64).reshape(-1))
+        out = np.full(len(nodes), -1, dtype=np.int64)
+        self.lib.ref_lrt_update(
+            len(nodes), nodes.ctypes.data_as(P64), len(src),
+            src.ctypes.data_as(P64), dst.ctypes.data_as(P64),
+            w.ctypes.data_as(PD) if weights is not None else None,
+            1 if directed else 0, mv.ctypes.data_as(P64), I64(len(mv)),
+            me.ctypes.data_as(P64), I64(len(me) // 2), dv.ctypes.data_as(P64),
+            I64(len(dv)), de.ctypes.data_as(P64), I64(len(de) // 2),
+            out.ctypes.data_as(P64))
+        return out
+
+    def reset(self):
+        self.lib.ref_lrt_reset()
+
+
+@pytest.fixture(scope="module")
+def nat():
+    n = Native()
+    if n.device_count() == 0:
+        pytest.skip("no HIP device")
+    return n
+
+
+@pytest.fixture(scope="module")
+def ctx(nat):
+    c = nat.init(0)
+    yield c
+    nat.destroy(c)
+
+
+@pytest.fixture
+def ref():
+    r = RefLrt()
+    r.reset()
+    yield r
+    r.reset()
+
+
+def make_graph(nat, ctx, V, src, dst, weights=None, directed=False):
+    if V == 0:
+        return None
+    if directed:
+        flags = BUILD_IN_CSR | BUILD_NO_PERM
+    else:
+        flags = BUILD_SYM_CSR
+    if weights is not None:
+        flags |= BUILD_WEIGHTED
+    return nat.graph_from_coo(ctx, src, dst, V, weights=weights, flags=flags)
+
+
+# two triangles + a bridge (the reference e2e community shape)
+V6 = 6
+SRC6 = [0, 1, 2, 3, 4, 5, 2]
+DST6 = [1, 2, 0, 4, 5, 3, 3]
+
+
+def test_lrt_set_undirected(nat, ctx, ref):
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V6, SRC6, DST6)
+    got = nat.lrt_set(ctx, g, list(range(V6)))
+    exp = ref.set(list(range(V6)), SRC6, DST6)
+    assert np.array_equal(got, exp), (got, exp)
+    # and two communities for the two triangles
+    assert got[0] == got[1] == got[2]
+    assert got[3] == got[4] == got[5]
+    assert got[0] != got[3]
+    nat.graph_destroy(ctx, g)
+
+
+def test_lrt_set_directed(nat, ctx, ref):
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V6, SRC6, DST6, directed=True)
+    got = nat.lrt_set(ctx, g, list(range(V6)), directed=True)
+    exp = ref.set(list(range(V6)), SRC6, DST6, directed=True)
+    assert np.array_equal(got, exp), (got, exp)
+    nat.graph_destroy(ctx, g)
+
+
+def test_lrt_set_weighted(nat, ctx, ref):
+    nat.lrt_reset(ctx)
+    w = [1.0, 2.0, 1.0, 0.5, 1.0, 2.0, 0.25]
+    g = make_graph(nat, ctx, V6, SRC6, DST6, weights=w)
+    got = nat.lrt_set(ctx, g, list(range(V6)), weighted=True, w_selfloop=0.5)
+    exp = ref.set(list(range(V6)), SRC6, DST6, weights=w, weighted=True,
+                  w_selfloop=0.5)
+    assert np.array_equal(got, exp), (got, exp)
+    nat.graph_destroy(ctx, g)
+
+
+def test_lrt_set_random(nat, ctx, ref):
+    rng = np.random.RandomState(5)
+    V, E = 60, 200
+    src = list(rng.randint(0, V, E))
+    dst = list(rng.randint(0, V, E))
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V, src, dst)
+    got = nat.lrt_set(ctx, g, list(range(V)))
+    exp = ref.set(list(range(V)), src, dst)
+    assert np.array_equal(got, exp), (got, exp)
+    nat.graph_destroy(ctx, g)
+
+
+def test_lrt_update_add_edge(nat, ctx, ref):
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V6, SRC6, DST6)
+    nat.lrt_set(ctx, g, list(range(V6)))
+    ref.set(list(range(V6)), SRC6, DST6)
+    nat.graph_destroy(ctx, g)
+    # add an edge 0-4 (cross-community)
+    src2 = SRC6 + [0]
+    dst2 = DST6 + [4]
+    g2 = make_graph(nat, ctx, V6, src2, dst2)
+    got = nat.lrt_update(ctx, g2, list(range(V6)), me=[(0, 4)])
+    exp = ref.update(list(range(V6)), src2, dst2, me=[(0, 4)])
+    assert np.array_equal(got, exp), (got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_lrt_update_add_vertex(nat, ctx, ref):
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V6, SRC6, DST6)
+    nat.lrt_set(ctx, g, list(range(V6)))
+    ref.set(list(range(V6)), SRC6, DST6)
+    nat.graph_destroy(ctx, g)
+    # new vertex 6 attached to the first triangle
+    src2 = SRC6 + [6, 0]
+    dst2 = DST6 + [0, 6]
+    g2 = make_graph(nat, ctx, 7, src2, dst2)
+    got = nat.lrt_update(ctx, g2, list(range(7)), mv=[6], me=[(6, 0), (0, 6)])
+    exp = ref.update(list(range(7)), src2, dst2, mv=[6], me=[(6, 0), (0, 6)])
+    assert np.array_equal(got, exp), (got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_lrt_update_delete_vertex(nat, ctx, ref):
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V6, SRC6, DST6)
+    nat.lrt_set(ctx, g, list(range(V6)))
+    ref.set(list(range(V6)), SRC6, DST6)
+    nat.graph_destroy(ctx, g)
+    # detach-delete vertex 5 (edges 4-5, 5-3)
+    keep = [i for i in range(len(SRC6)) if SRC6[i] != 5 and DST6[i] != 5]
+    src2 = [SRC6[i] for i in keep]
+    dst2 = [DST6[i] for i in keep]
+    # remaining graph has nodes 0..4; dense ids still 0..4
+    g2 = make_graph(nat, ctx, 5, src2, dst2)
+    got = nat.lrt_update(ctx, g2, list(range(5)), dv=[5], de=[(4, 5), (5, 3)])
+    exp = ref.update(list(range(5)), src2, dst2, dv=[5], de=[(4, 5), (5, 3)])
+    assert np.array_equal(got, exp), (got, exp)
+    nat.graph_destroy(ctx, g2)
+
+
+def test_lrt_get_semantics(nat, ctx, ref):
+    """get on uncalculated state runs a full non-persisted compute
+    (GetLabels :305-309)."""
+    nat.lrt_reset(ctx)
+    g = make_graph(nat, ctx, V6, SRC6, DST6)
+    got, ran = nat.lrt_get(ctx, g, list(range(V6)))
+    assert ran == 1
+    ref.reset()
+    # GetLabels defaults == SetLabels defaults
+    exp = ref.set(list(range(V6)), SRC6, DST6)
+    assert np.array_equal(got, exp)
+    nat.graph_destroy(ctx, g)

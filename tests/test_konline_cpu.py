@@ -222,3 +222,37 @@ def test_katz_online_module_registration():
         ["created_vertices", "nullable"], ["created_edges", "nullable"],
         ["deleted_vertices", "nullable"], ["deleted_edges", "nullable"]]
     assert procs["reset"]["results"] == [["message", "string"]]
+
+
+def test_cd_online_module_registration():
+    """community_detection_online.so registers the reference's exact
+    procedures (community_detection_online_module.cpp:228-325)."""
+    import subprocess
+    import sys as _sys
+    mockdir = os.path.join(os.path.dirname(__file__), "mock")
+    modules = os.path.join(REPO, "memgraph_amd", "lib", "modules")
+    if not os.path.exists(os.path.join(modules, "community_detection_online.so")):
+        pytest.skip("community_detection_online.so not built")
+    script = (
+        "import sys, json\n"
+        f"sys.path.insert(0, {mockdir!r})\n"
+        "from harness import ModuleHost\n"
+        "h = ModuleHost('community_detection_online')\n"
+        "print(json.dumps(h.procedures()))\n")
+    out = subprocess.run([_sys.executable, "-c", script], capture_output=True,
+                         text=True)
+    assert out.returncode == 0, out.stderr
+    import json
+    procs = json.loads(out.stdout)
+    assert procs["set"]["args"] == [
+        ["directed", "bool"], ["weighted", "bool"],
+        ["similarity_threshold", "float"], ["exponent", "float"],
+        ["min_value", "float"], ["weight_property", "string"],
+        ["w_selfloop", "float"], ["max_iterations", "int"],
+        ["max_updates", "int"]]
+    assert procs["set"]["results"] == [["node", "node"], ["community_id", "int"]]
+    assert procs["update"]["args"] == [
+        ["createdVertices", "nullable"], ["createdEdges", "nullable"],
+        ["updatedVertices", "nullable"], ["updatedEdges", "nullable"],
+        ["deletedVertices", "nullable"], ["deletedEdges", "nullable"]]
+    assert procs["reset"]["results"] == [["message", "string"]]
