@@ -560,8 +560,9 @@ extern "C" int64_t mgx_konline_iterations(void) { return g_k.iteration; }
 extern "C" mgx_status mgx_konline_set(mgx_context *ctx, mgx_graph *g,
                                       const int64_t *dense_to_mg, double alpha,
                                       double epsilon, double *out) {
-  if (g && !((g->flags & MGX_BUILD_IN_CSR) && (g->flags & MGX_BUILD_OUT_CSR))) {
-    mgx_set_error("konline_set needs IN_CSR|OUT_CSR");
+  if (g && !((g->flags & MGX_BUILD_IN_CSR) && (g->flags & MGX_BUILD_OUT_CSR) &&
+             (g->flags & MGX_BUILD_NO_PERM))) {
+    mgx_set_error("konline_set needs IN_CSR|OUT_CSR|NO_PERM (identity layout)");
     return MGX_ERR_INVALID_ARGUMENT;
   }
   MGX_HIP_TRY(hipSetDevice(ctx->device));
@@ -620,8 +621,9 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
                                          const int64_t *deleted_v, int64_t n_dv,
                                          const int64_t *deleted_e, int64_t n_de,
                                          double *out) {
-  if (g && !((g->flags & MGX_BUILD_IN_CSR) && (g->flags & MGX_BUILD_OUT_CSR))) {
-    mgx_set_error("konline_update needs IN_CSR|OUT_CSR");
+  if (g && !((g->flags & MGX_BUILD_IN_CSR) && (g->flags & MGX_BUILD_OUT_CSR) &&
+             (g->flags & MGX_BUILD_NO_PERM))) {
+    mgx_set_error("konline_update needs IN_CSR|OUT_CSR|NO_PERM (identity layout)");
     return MGX_ERR_INVALID_ARGUMENT;
   }
   MGX_HIP_TRY(hipSetDevice(ctx->device));

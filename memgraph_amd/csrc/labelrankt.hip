@@ -847,6 +847,20 @@ extern "C" mgx_status mgx_lrt_reset(mgx_context *ctx) {
   return MGX_OK;
 }
 
+static mgx_status lrt_check_graph(mgx_graph *g, bool directed) {
+  if (!g) return MGX_OK;
+  if (directed) {
+    if (!((g->flags & MGX_BUILD_IN_CSR) && (g->flags & MGX_BUILD_NO_PERM))) {
+      mgx_set_error("lrt directed needs IN_CSR|NO_PERM");
+      return MGX_ERR_INVALID_ARGUMENT;
+    }
+  } else if (!(g->flags & MGX_BUILD_SYM_CSR)) {
+    mgx_set_error("lrt undirected needs SYM_CSR");
+    return MGX_ERR_INVALID_ARGUMENT;
+  }
+  return MGX_OK;
+}
+
 extern "C" mgx_status mgx_lrt_set(mgx_context *ctx, mgx_graph *g,
                                   const int64_t *dense_to_mg, int directed, int weighted,
                                   double similarity_threshold, double exponent,
@@ -854,6 +868,7 @@ extern "C" mgx_status mgx_lrt_set(mgx_context *ctx, mgx_graph *g,
                                   int64_t max_iterations, int64_t max_updates,
                                   int64_t *out_label) {
   MGX_HIP_TRY(hipSetDevice(ctx->device));
+  MGX_TRY(lrt_check_graph(g, directed != 0));
   lrt_free();
   g_l.directed = directed != 0;
   g_l.weighted = weighted != 0;

@@ -55,7 +55,8 @@ struct OnlineCall {
     if (V > 0) {
       CheckMgx(mgx_graph_from_coo(gg.ctx, scan.src.data(), scan.dst.data(), nullptr, V,
                                   (int64_t)scan.src.size(),
-                                  MGX_BUILD_IN_CSR | MGX_BUILD_OUT_CSR, &gg.g),
+                                  MGX_BUILD_IN_CSR | MGX_BUILD_OUT_CSR | MGX_BUILD_NO_PERM,
+                                  &gg.g),
                "mgx_graph_from_coo");
     }
     rank.resize(V);

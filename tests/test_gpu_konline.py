@@ -14,13 +14,14 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
 sys.path.insert(0, os.path.dirname(__file__))
 
-from memgraph_amd.native import BUILD_IN_CSR, BUILD_OUT_CSR, Native  # noqa: E402
+from memgraph_amd.native import (BUILD_IN_CSR, BUILD_NO_PERM, BUILD_OUT_CSR,
+                                 Native)  # noqa: E402
 from test_konline_cpu import KOracle  # noqa: E402
 import ctypes  # noqa: E402
 
 pytestmark = pytest.mark.gpu
 
-FLAGS = BUILD_IN_CSR | BUILD_OUT_CSR
+FLAGS = BUILD_IN_CSR | BUILD_OUT_CSR | BUILD_NO_PERM  # identity layout
 
 
 @pytest.fixture(scope="module")
