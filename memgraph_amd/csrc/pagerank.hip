@@ -45,6 +45,7 @@ struct PrArgs {
   const uint32_t *sp_hi;
   double *partial;
   int mode;
+  int deep;  // experiment: 4x-unrolled wide path (16 gathers/lane in flight)
 };
 
 template <int LANES>
@@ -74,8 +75,28 @@ __device__ inline float pr_rows(const PrArgs &A, int sec, int64_t block_in_sec) 
         const uint32_t nvec = (e - s_al) / 4;
         typedef int v4i __attribute__((ext_vector_type(4)));
         const v4i *col4 = reinterpret_cast<const v4i *>(A.col + s_al);
-        // 2x unrolled: 8 independent gathers in flight per lane.
         uint32_t c = sub;
+        if (A.deep) {
+          // 4x unrolled: 16 independent gathers in flight per lane
+          // (MGX_PR_DEEP_UNROLL experiment).
+          for (; c + 3 * LANES < nvec; c += 4 * LANES) {
+            const v4i c0 = __builtin_nontemporal_load(col4 + c);
+            const v4i c1 = __builtin_nontemporal_load(col4 + c + LANES);
+            const v4i c2 = __builtin_nontemporal_load(col4 + c + 2 * LANES);
+            const v4i c3 = __builtin_nontemporal_load(col4 + c + 3 * LANES);
+            float g[16];
+            g[0] = A.contrib_old[c0.x]; g[1] = A.contrib_old[c0.y];
+            g[2] = A.contrib_old[c0.z]; g[3] = A.contrib_old[c0.w];
+            g[4] = A.contrib_old[c1.x]; g[5] = A.contrib_old[c1.y];
+            g[6] = A.contrib_old[c1.z]; g[7] = A.contrib_old[c1.w];
+            g[8] = A.contrib_old[c2.x]; g[9] = A.contrib_old[c2.y];
+            g[10] = A.contrib_old[c2.z]; g[11] = A.contrib_old[c2.w];
+            g[12] = A.contrib_old[c3.x]; g[13] = A.contrib_old[c3.y];
+            g[14] = A.contrib_old[c3.z]; g[15] = A.contrib_old[c3.w];
+            for (int k = 0; k < 16; ++k) acc += (double)g[k];
+          }
+        }
+        // 2x unrolled: 8 independent gathers in flight per lane.
         for (; c + LANES < nvec; c += 2 * LANES) {
           const v4i c0 = __builtin_nontemporal_load(col4 + c);
           const v4i c1 = __builtin_nontemporal_load(col4 + c + LANES);
@@ -231,6 +252,13 @@ mgx_status queue_one_iteration(mgx_pagerank_run *run, bool track_delta) {
   A.damping = (float)run->damping;
   A.delta_max = track_delta ? run->d_delta : nullptr;
   A.partial = run->d_partial;
+  {
+    static const int deep = [] {
+      const char *e = getenv("MGX_PR_DEEP_UNROLL");
+      return e && atoi(e) ? 1 : 0;
+    }();
+    A.deep = deep;
+  }
 
   if (track_delta) MGX_HIP_TRY(hipMemsetAsync(run->d_delta, 0, 4, ctx->stream));
 
