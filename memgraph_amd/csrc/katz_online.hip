@@ -419,11 +419,13 @@ mgx_status kon_build_maps(mgx_context *ctx, mgx_graph *g, const int64_t *dense_t
     m->n_big = (int64_t)bg.size();
     MGX_HIP_TRY(hipMalloc(&m->small_rows, (m->n_small > 0 ? m->n_small : 1) * 4));
     MGX_HIP_TRY(hipMalloc(&m->big_rows, (m->n_big > 0 ? m->n_big : 1) * 4));
-    MGX_HIP_TRY(hipMemcpyAsync(m->small_rows, sm.data(),
-                               (m->n_small > 0 ? m->n_small : 1) * 4,
-                               hipMemcpyHostToDevice, ctx->stream));
-    MGX_HIP_TRY(hipMemcpyAsync(m->big_rows, bg.data(), (m->n_big > 0 ? m->n_big : 1) * 4,
-                               hipMemcpyHostToDevice, ctx->stream));
+    // empty vectors have a null data(): copy only when non-empty
+    if (m->n_small > 0)
+      MGX_HIP_TRY(hipMemcpyAsync(m->small_rows, sm.data(), m->n_small * 4,
+                                 hipMemcpyHostToDevice, ctx->stream));
+    if (m->n_big > 0)
+      MGX_HIP_TRY(hipMemcpyAsync(m->big_rows, bg.data(), m->n_big * 4,
+                                 hipMemcpyHostToDevice, ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   }
   return MGX_OK;
@@ -446,8 +448,9 @@ mgx_status kon_loop(mgx_context *ctx, mgx_graph *g, const KMaps &m, double gamma
   MGX_HIP_TRY(hipMalloc(&d_keys, (n_active > 0 ? n_active : 1) * 8));
   MGX_HIP_TRY(hipMalloc(&d_keys_sorted, (n_active > 0 ? n_active : 1) * 8));
   MGX_HIP_TRY(hipMalloc(&d_flag, 4));
-  MGX_HIP_TRY(hipMemcpyAsync(d_act, act.data(), (n_active > 0 ? n_active : 1) * 4,
-                             hipMemcpyHostToDevice, ctx->stream));
+  if (n_active > 0)
+    MGX_HIP_TRY(hipMemcpyAsync(d_act, act.data(), n_active * 4, hipMemcpyHostToDevice,
+                               ctx->stream));
 
   mgx_status st = MGX_OK;
   while (true) {

@@ -20,6 +20,8 @@ void mgx_set_error(const char *fmt, ...);
     if (_e != hipSuccess) {                                                   \
       mgx_set_error("%s:%d: %s failed: %s", __FILE__, __LINE__, #expr,        \
                     hipGetErrorString(_e));                                   \
+      (void)hipGetLastError(); /* consume the sticky per-thread error so a  \
+                                  later rocprim dispatch doesn't inherit it */ \
       return MGX_ERR_HIP;                                                     \
     }                                                                         \
   } while (0)
