@@ -11,13 +11,13 @@ LIBDIR := memgraph_amd/lib
 MGX_SRCS := memgraph_amd/csrc/mgx_api.cpp memgraph_amd/csrc/comm.cpp
 MGX_HIP_SRCS := memgraph_amd/csrc/graph_build.hip memgraph_amd/csrc/pagerank.hip \
                 memgraph_amd/csrc/pronline.hip memgraph_amd/csrc/katz_online.hip \
-                memgraph_amd/csrc/labelrankt.hip \
+                memgraph_amd/csrc/labelrankt.hip memgraph_amd/csrc/leiden.hip \
                 memgraph_amd/csrc/wcc.hip memgraph_amd/csrc/katz.hip \
                 memgraph_amd/csrc/betweenness.hip \
                 memgraph_amd/csrc/louvain.hip
 MGX_OBJS := $(MGX_SRCS:%.cpp=build/%.o) $(MGX_HIP_SRCS:%.hip=build/%.o)
 
-MODULES := pagerank katz_centrality community_detection weakly_connected_components betweenness_centrality pagerank_online katz_centrality_online community_detection_online
+MODULES := pagerank katz_centrality community_detection weakly_connected_components betweenness_centrality pagerank_online katz_centrality_online community_detection_online leiden_community_detection
 
 all: $(LIBDIR)/libmgx_analytics.so modules mock oracle
 

@@ -245,6 +245,20 @@ mgx_status mgx_lrt_update(mgx_context *ctx, mgx_graph *g, const int64_t *dense_t
 mgx_status mgx_lrt_reset(mgx_context *ctx);
 int mgx_lrt_initialized(void);
 
+/* ---- Leiden community detection — replaces leiden_alg::GetCommunities
+ *      (src/mage/cpp/leiden_community_detection_module/algorithm/
+ *      leiden.cpp:569-591; CPM local moves, probabilistic refinement,
+ *      first-edge aggregation). RANDOMIZED in the reference
+ *      (random_device-seeded shuffle/draws), so parity is the DESIGN.md
+ *      statistical bar; `seed` pins our counter-based streams.
+ * Graph needs MGX_BUILD_SYM_CSR (|WEIGHTED). out_hier[v*cap+k] = the
+ * community id of v at dendrogram level k (-1 padded); out_levels[v] =
+ * hierarchy depth (0 when the graph has no edges — the module then raises
+ * the reference's "No communities detected." error path). */
+mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma, double theta,
+                      double resolution, int64_t max_iterations, uint64_t seed,
+                      int64_t cap, int64_t *out_hier, int64_t *out_levels);
+
 /* ---- multi-GPU (RCCL over xGMI) ------------------------- */
 
 #define MGX_UNIQUE_ID_BYTES 128 /* == sizeof(ncclUniqueId) */

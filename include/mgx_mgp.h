@@ -97,6 +97,8 @@ enum mgp_error mgp_list_make_empty(size_t capacity, struct mgp_memory *memory,
                                    struct mgp_list **result);               /* :519 */
 enum mgp_error mgp_value_make_list(struct mgp_list *val,
                                    struct mgp_value **result);              /* :256 */
+enum mgp_error mgp_list_append_extend(struct mgp_list *list,
+                                      struct mgp_value *val);               /* :544 */
 enum mgp_error mgp_list_size(struct mgp_list *list, size_t *result);        /* :554 */
 enum mgp_error mgp_list_at(struct mgp_list *list, size_t index,
                            struct mgp_value **result);                      /* :563 */

@@ -281,6 +281,22 @@ inline void InsertString(mgp_result_record *record, const char *field, const cha
   mgp_value_destroy(val);
 }
 
+inline void InsertIntList(mgp_result_record *record, const char *field,
+                          const std::vector<int64_t> &vals, mgp_memory *memory) {
+  mgp_list *list = nullptr;
+  Check(mgp_list_make_empty(vals.size(), memory, &list), "list_make_empty");
+  for (int64_t v : vals) {
+    mgp_value *iv = nullptr;
+    Check(mgp_value_make_int(v, memory, &iv), "value_make_int");
+    Check(mgp_list_append_extend(list, iv), "list_append_extend");
+    mgp_value_destroy(iv);
+  }
+  mgp_value *lv = nullptr;
+  Check(mgp_value_make_list(list, &lv), "value_make_list");
+  Check(mgp_result_record_insert(record, field, lv), "record_insert(list)");
+  mgp_value_destroy(lv);
+}
+
 // Argument readers (procedure-supplied args arrive positionally).
 inline int64_t ArgInt(mgp_list *args, size_t i) {
   mgp_value *v = nullptr;

@@ -355,6 +355,20 @@ class Native:
     def lrt_reset(self, ctx):
         self._check(self.lib.mgx_lrt_reset(ctx), "mgx_lrt_reset")
 
+    # --- leiden ---
+    def leiden(self, ctx, g, n_vertices, gamma=1.0, theta=0.01, resolution=0.01,
+               max_iterations=(1 << 62), seed=1, cap=64):
+        hier = np.full(n_vertices * cap, -1, dtype=np.int64)
+        levels = np.zeros(n_vertices, dtype=np.int64)
+        self._check(
+            self.lib.mgx_leiden(ctx, g, ctypes.c_double(gamma), ctypes.c_double(theta),
+                                ctypes.c_double(resolution),
+                                ctypes.c_int64(max_iterations), ctypes.c_uint64(seed),
+                                ctypes.c_int64(cap), hier.ctypes.data_as(_I64),
+                                levels.ctypes.data_as(_I64)),
+            "mgx_leiden")
+        return hier.reshape(n_vertices, cap), levels
+
     # --- comm ---
     def comm_unique_id(self):
         buf = (ctypes.c_char * UNIQUE_ID_BYTES)()

@@ -200,3 +200,35 @@ void ref_lrt_update(int64_t n_nodes, const int64_t *nodes, int64_t n_edges,
 }
 
 }  // extern "C"
+
+// ---- Leiden (randomized: distribution-level pin; goldens are stable) -----
+// Drives leiden_alg::GetCommunities (src/mage/cpp/
+// leiden_community_detection_module/algorithm/leiden.cpp:569-591) compiled
+// from the reference sources (boost replaced by the std shim in shim/boost).
+#include "leiden.hpp"
+
+extern "C" void ref_leiden(int64_t n_nodes, int64_t n_edges, const int64_t *src,
+                           const int64_t *dst, const double *weights, double gamma,
+                           double theta, double resolution, int64_t max_iterations,
+                           int64_t cap, int64_t *out_hier /* [n_nodes*cap], -1 pad */,
+                           int64_t *out_levels /* [n_nodes] */) {
+  mg_graph::Graph<> g;
+  for (int64_t v = 0; v < n_nodes; ++v) g.CreateNode((uint64_t)v);
+  for (int64_t e = 0; e < n_edges; ++e) {
+    g.CreateEdge((uint64_t)src[e], (uint64_t)dst[e], mg_graph::GraphType::kUndirectedGraph,
+                 std::nullopt, weights != nullptr, weights ? weights[e] : 0.0);
+  }
+  auto hier = leiden_alg::GetCommunities(g, gamma, theta, resolution,
+                                         (uint64_t)max_iterations);
+  for (int64_t v = 0; v < n_nodes; ++v) {
+    int64_t L = 0;
+    if ((size_t)v < hier.size()) {
+      for (auto c : hier[v]) {
+        if (L < cap) out_hier[v * cap + L] = (int64_t)c;
+        ++L;
+      }
+    }
+    for (int64_t k = L; k < cap; ++k) out_hier[v * cap + k] = -1;
+    out_levels[v] = L;
+  }
+}

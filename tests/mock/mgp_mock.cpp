@@ -306,6 +306,16 @@ int mgp_is_enterprise_valid(void) {
   return (e && e[0] == '0') ? 0 : 1;
 }
 
+enum mgp_error mgp_list_append_extend(struct mgp_list *list, struct mgp_value *val) {
+  // mg_procedure.h:544: append, extending capacity as needed; the mock
+  // copies the value into the process arena so result lists stay valid.
+  auto v = std::make_unique<mgp_value>();
+  v->v = val->v;
+  list->items.push_back(v.get());
+  g_arena_values.push_back(std::move(v));
+  return MGP_ERROR_NO_ERROR;
+}
+
 enum mgp_error mgp_list_size(struct mgp_list *list, size_t *result) {
   *result = list->items.size();
   return MGP_ERROR_NO_ERROR;
@@ -722,6 +732,17 @@ double mock_result_double(int64_t row, const char *field) {
 const char *mock_result_string(int64_t row, const char *field) {
   auto &f = g_result.rows[row].fields.at(field);
   return f.s.c_str();
+}
+
+int64_t mock_result_list_len(int64_t row, const char *field) {
+  auto &f = g_result.rows[row].fields.at(field);
+  if (!f.list) return -1;
+  return (int64_t)f.list->items.size();
+}
+
+int64_t mock_result_list_int(int64_t row, const char *field, int64_t i) {
+  auto &f = g_result.rows[row].fields.at(field);
+  return f.list->items[i]->v.i;
 }
 
 }  // extern "C"
