@@ -218,8 +218,21 @@ extern "C" void ref_leiden(int64_t n_nodes, int64_t n_edges, const int64_t *src,
     g.CreateEdge((uint64_t)src[e], (uint64_t)dst[e], mg_graph::GraphType::kUndirectedGraph,
                  std::nullopt, weights != nullptr, weights ? weights[e] : 0.0);
   }
-  auto hier = leiden_alg::GetCommunities(g, gamma, theta, resolution,
-                                         (uint64_t)max_iterations);
+  std::vector<std::vector<uint64_t>> hier;
+  try {
+    hier = leiden_alg::GetCommunities(g, gamma, theta, resolution,
+                                      (uint64_t)max_iterations);
+  } catch (const std::exception &) {
+    // the reference throws "No communities detected." when its (randomized)
+    // trajectory finishes without an aggregation level (leiden.cpp:585-586)
+    // — surface as levels = -1 so callers can treat the run as a
+    // reference-error outcome
+    for (int64_t v = 0; v < n_nodes; ++v) {
+      out_levels[v] = -1;
+      for (int64_t k = 0; k < cap; ++k) out_hier[v * cap + k] = -1;
+    }
+    return;
+  }
   for (int64_t v = 0; v < n_nodes; ++v) {
     int64_t L = 0;
     if ((size_t)v < hier.size()) {
