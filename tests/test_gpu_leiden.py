@@ -136,41 +136,54 @@ def _cpm_quality(V, src, dst, labels, gamma_norm):
 
 
 def test_leiden_quality_random(nat, ctx):
-    """Random graph: GPU partition quality (CPM) within the reference's own
-    run spread (min - tolerance), community count within range."""
+    """Planted-partition graph (8 dense clusters, sparse bridges): CPM finds
+    the planted structure robustly in both implementations, so community
+    count and CPM quality must sit inside the reference's run spread. A
+    uniform random graph is NOT used here: on those, even the reference's
+    own runs swing between 3 and 40 communities and sometimes end in its
+    "No communities detected." error path (observed via ref_leiden), so no
+    cross-implementation bound is meaningful."""
     rng = np.random.RandomState(9)
-    V, E = 120, 480
-    src = list(rng.randint(0, V, E))
-    dst = list(rng.randint(0, V, E))
-    # unique undirected edge count for gamma normalization
+    K, M = 8, 15          # 8 clusters of 15 nodes
+    V = K * M
+    src, dst = [], []
+    for k in range(K):
+        base = k * M
+        for i in range(M):
+            for j in range(i + 1, M):
+                if rng.rand() < 0.5:
+                    src.append(base + i)
+                    dst.append(base + j)
+    for _ in range(20):   # sparse inter-cluster bridges
+        a, b = rng.randint(0, V), rng.randint(0, V)
+        if a // M != b // M:
+            src.append(a)
+            dst.append(b)
     uniq = {(min(s, d), max(s, d)) for s, d in zip(src, dst)}
     gamma_norm = 1.0 / sum(1.0 for _ in uniq)
 
     ref = RefLeiden()
     ref_q = []
     ref_k = []
-    for _ in range(14):
+    for _ in range(10):
         h, lv = ref.run(V, src, dst)
-        if (lv < 0).any():
-            # the reference's randomized trajectory threw "No communities
-            # detected." (merge-everything path, leiden.cpp:585-586): a
-            # reference-error outcome, not a sample
-            continue
-        labels = [h[v][lv[v] - 1] if lv[v] > 0 else -v - 1 for v in range(V)]
+        if (lv <= 0).any():
+            continue  # reference-error run (leiden.cpp:585-586)
+        labels = [h[v][lv[v] - 1] for v in range(V)]
         ref_q.append(_cpm_quality(V, src, dst, labels, gamma_norm))
         ref_k.append(len(set(labels)))
-    if len(ref_q) < 3:
-        pytest.skip("reference errored on most runs for this graph")
+    assert len(ref_q) >= 3, "reference errored on most planted-partition runs"
 
     g = nat.graph_from_coo(ctx, src, dst, V, flags=BUILD_SYM_CSR)
     gh, glv = nat.leiden(ctx, g, V, seed=7, cap=CAP)
-    labels = [gh[v][glv[v] - 1] if glv[v] > 0 else -v - 1 for v in range(V)]
+    assert (glv > 0).all()
+    labels = [gh[v][glv[v] - 1] for v in range(V)]
     q = _cpm_quality(V, src, dst, labels, gamma_norm)
     k = len(set(labels))
     spread = max(ref_q) - min(ref_q) + 1e-9
     assert q >= min(ref_q) - max(3 * spread, 0.05 * abs(min(ref_q)) + 1e-6), \
         (q, ref_q)
-    assert min(ref_k) // 2 - 2 <= k <= max(ref_k) * 2 + 2, (k, ref_k)
+    assert min(ref_k) - 2 <= k <= max(ref_k) + 2, (k, ref_k)
     nat.graph_destroy(ctx, g)
 
 
