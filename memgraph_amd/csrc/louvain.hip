@@ -43,6 +43,28 @@ constexpr int kBlock = 256;
 constexpr uint32_t kSmallRowDeg = 256;  // <: wave+LDS table; >=: block+pool
 constexpr int kLdsCap = 512;            // per-wave table entries (>= 2*255)
 
+inline bool louvain_trace() {
+  // MGX_LOUVAIN_TRACE: bare stderr phase markers with NO syncs and NO
+  // allocations — under AMD_SERIALIZE_KERNEL the process aborts inside the
+  // faulting launch, so the last marker identifies the kernel without
+  // perturbing timing or the allocation layout (the full DEBUG mode's
+  // extra buffers/syncs mask the storm fault — r02 finding).
+  static const bool v = [] {
+    const char *e = getenv("MGX_LOUVAIN_TRACE");
+    return e && atoi(e) != 0;
+  }();
+  return v;
+}
+
+#define MGX_LTRACE(...)                                                       \
+  do {                                                                        \
+    if (louvain_trace()) {                                                    \
+      fprintf(stderr, "[ltrace] " __VA_ARGS__);                               \
+      fprintf(stderr, "\n");                                                  \
+      fflush(stderr);                                                         \
+    }                                                                         \
+  } while (0)
+
 inline bool louvain_debug() {
   static const bool v = [] {
     const char *e = getenv("MGX_LOUVAIN_DEBUG");
@@ -738,6 +760,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     fflush(stderr);
   }
   // Binned row weight sums (sumVertexDegree) over the just-built lists.
+  MGX_LTRACE("setup k_row_wsum_init nv=%lld", (long long)nv);
   hipLaunchKernelGGL(k_row_wsum_init, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                      ctx->stream, nv, vdeg.as<double>(), cinfo_deg.as<double>(),
                      cinfo_size.as<int32_t>());
@@ -790,6 +813,8 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   }
 
   MGX_LDBG(ctx, "pool_total=%llu", (unsigned long long)pool_total);
+  MGX_LTRACE("setup done pool_total=%llu n_small=%lld n_big=%lld",
+             (unsigned long long)pool_total, (long long)n_small, (long long)n_big);
   SweepArgs A;
   A.row_ptr = L.row_ptr;
   A.col = L.col;
@@ -835,29 +860,34 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     A.curr = p_curr;
     A.target = p_target;
     MGX_LDBG(ctx, "iter=%lld pre-sweep", (long long)iters);
+    MGX_LTRACE("it=%lld k_sweep_empty", (long long)iters);
     hipLaunchKernelGGL(k_sweep_empty, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, L.row_ptr, p_target);
     MGX_LDBG(ctx, "iter=%lld after-empty", (long long)iters);
     if (n_small > 0) {
       const int64_t waves_needed = n_small;
       const int64_t blocks = grid_for(waves_needed * 64, 4096);
+      MGX_LTRACE("it=%lld k_sweep_small", (long long)iters);
       hipLaunchKernelGGL(k_sweep_small, dim3((uint32_t)blocks), dim3(kBlock), 0,
                          ctx->stream, A);
     }
     MGX_LDBG(ctx, "iter=%lld after-small", (long long)iters);
     if (n_big > 0) {
       if (!A.inkernel_clear) {
+        MGX_LTRACE("it=%lld pool_clear", (long long)iters);
         hipLaunchKernelGGL(k_fill_i32, dim3((uint32_t)grid_for((int64_t)pool_total)),
                            dim3(kBlock), 0, ctx->stream, (int64_t)pool_total, -1,
                            pool_keys.as<int32_t>());
         MGX_HIP_TRY(hipMemsetAsync(pool_vals.p, 0, pool_total * 8, ctx->stream));
       }
       const int64_t blocks = n_big < 4096 ? n_big : 4096;
+      MGX_LTRACE("it=%lld k_sweep_big", (long long)iters);
       hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
                          A);
     }
     MGX_LDBG(ctx, "iter=%lld after-big", (long long)iters);
     MGX_LDBG(ctx, "iter=%lld post-sweep", (long long)iters);
+    MGX_LTRACE("it=%lld k_sum_sq", (long long)iters);
     hipLaunchKernelGGL(k_sum_sq_f64, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, cinfo_deg.as<double>(), scalars.as<double>() + 1);
     double exx_a2x[2] = {0.0, 0.0};
@@ -872,6 +902,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     if ((curr_mod - prev_mod) < thresh) break;
     prev_mod = curr_mod;
     if (prev_mod < lower) prev_mod = lower;
+    MGX_LTRACE("it=%lld k_apply_updates", (long long)iters);
     hipLaunchKernelGGL(k_apply_updates, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                        ctx->stream, nv, cinfo_deg.as<double>(), cinfo_size.as<int32_t>(),
                        cupd_deg.as<double>(), cupd_size.as<int32_t>());
