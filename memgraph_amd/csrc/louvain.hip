@@ -322,8 +322,13 @@ __global__ void __launch_bounds__(kBlock) k_sweep_small(SweepArgs A) {
     double own = 0.0;
     {
       uint32_t h = ((uint32_t)sc * 2654435761u) & (kLdsCap - 1);
-      while (keys[wave][h] != sc) h = (h + 1) & (kLdsCap - 1);
-      own = vals[wave][h];
+      int probes = 0;
+      while (keys[wave][h] != sc && ++probes <= kLdsCap) h = (h + 1) & (kLdsCap - 1);
+      if (probes > kLdsCap) {
+        atomicOr(&A.guard[0], 16u);  // own-community entry vanished from LDS
+      } else {
+        own = vals[wave][h];
+      }
     }
     const double eix = own - self_loop;
     const double ax = A.cinfo_deg[sc] - A.vdeg[row];
@@ -417,8 +422,13 @@ __global__ void __launch_bounds__(kBlock) k_sweep_big(SweepArgs A) {
     double own = 0.0;
     {
       uint32_t h = ((uint32_t)sc * 2654435761u) & (cap - 1);
-      while (keys[h] != sc) h = (h + 1) & (cap - 1);
-      own = vals[h];
+      uint32_t probes = 0;
+      while (keys[h] != sc && ++probes <= cap) h = (h + 1) & (cap - 1);
+      if (probes > cap) {
+        atomicOr(&A.guard[0], 32u);  // own-community entry vanished from pool
+      } else {
+        own = vals[h];
+      }
     }
     const double eix = own - sl;
     const double ax = A.cinfo_deg[sc] - A.vdeg[row];
@@ -870,6 +880,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
       MGX_LTRACE("it=%lld k_sweep_small", (long long)iters);
       hipLaunchKernelGGL(k_sweep_small, dim3((uint32_t)blocks), dim3(kBlock), 0,
                          ctx->stream, A);
+      MGX_LTRACE("it=%lld k_sweep_small launched", (long long)iters);
     }
     MGX_LDBG(ctx, "iter=%lld after-small", (long long)iters);
     if (n_big > 0) {
@@ -884,6 +895,7 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
       MGX_LTRACE("it=%lld k_sweep_big", (long long)iters);
       hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
                          A);
+      MGX_LTRACE("it=%lld k_sweep_big launched", (long long)iters);
     }
     MGX_LDBG(ctx, "iter=%lld after-big", (long long)iters);
     MGX_LDBG(ctx, "iter=%lld post-sweep", (long long)iters);
@@ -893,6 +905,14 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
     double exx_a2x[2] = {0.0, 0.0};
     MGX_HIP_TRY(hipMemcpyAsync(exx_a2x, scalars.p, 16, hipMemcpyDeviceToHost, ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    if (louvain_trace()) {
+      uint32_t ge[4] = {0, 0, 0, 0};
+      MGX_HIP_TRY(hipMemcpyAsync(ge, A.guard, 16, hipMemcpyDeviceToHost, ctx->stream));
+      MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+      if (ge[0] || ge[1])
+        MGX_LTRACE("it=%lld GUARD bits=%u oob=%u row=%u cid=%u", (long long)iters, ge[0],
+                   ge[1], ge[2], ge[3]);
+    }
     curr_mod = exx_a2x[0] * constant - exx_a2x[1] * constant * constant;
     if (louvain_debug() && (iters < 10 || iters % 500 == 0)) {
       fprintf(stderr, "[louvain] iter=%lld exx=%.12g a2x=%.12g mod=%.12g prev=%.12g\n",
