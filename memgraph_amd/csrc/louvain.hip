@@ -65,6 +65,15 @@ inline bool louvain_trace() {
     }                                                                         \
   } while (0)
 
+inline int louvain_diag(const char *name) {
+  // diagnostic-only switches for the storm-fault bisect (r02):
+  //   MGX_LOUVAIN_SKIP_SMALL / _SKIP_BIG: omit a sweep kernel (results
+  //   WRONG — bisect only), MGX_LOUVAIN_SYNC_AFTER: sync+getLastError
+  //   after each sweep launch and print the status.
+  const char *e = getenv(name);
+  return e && atoi(e) != 0;
+}
+
 inline bool louvain_debug() {
   static const bool v = [] {
     const char *e = getenv("MGX_LOUVAIN_DEBUG");
@@ -878,8 +887,18 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
       const int64_t waves_needed = n_small;
       const int64_t blocks = grid_for(waves_needed * 64, 4096);
       MGX_LTRACE("it=%lld k_sweep_small", (long long)iters);
-      hipLaunchKernelGGL(k_sweep_small, dim3((uint32_t)blocks), dim3(kBlock), 0,
-                         ctx->stream, A);
+      static const int skip_small = louvain_diag("MGX_LOUVAIN_SKIP_SMALL");
+      if (!skip_small)
+        hipLaunchKernelGGL(k_sweep_small, dim3((uint32_t)blocks), dim3(kBlock), 0,
+                           ctx->stream, A);
+      static const int sync_after = louvain_diag("MGX_LOUVAIN_SYNC_AFTER");
+      if (sync_after) {
+        hipError_t se = hipStreamSynchronize(ctx->stream);
+        hipError_t le = hipGetLastError();
+        if (se != hipSuccess || le != hipSuccess)
+          MGX_LTRACE("it=%lld SMALL sync=%s last=%s", (long long)iters,
+                     hipGetErrorString(se), hipGetErrorString(le));
+      }
       MGX_LTRACE("it=%lld k_sweep_small launched", (long long)iters);
     }
     MGX_LDBG(ctx, "iter=%lld after-small", (long long)iters);
@@ -893,8 +912,18 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
       }
       const int64_t blocks = n_big < 4096 ? n_big : 4096;
       MGX_LTRACE("it=%lld k_sweep_big", (long long)iters);
-      hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0, ctx->stream,
-                         A);
+      static const int skip_big = louvain_diag("MGX_LOUVAIN_SKIP_BIG");
+      if (!skip_big)
+        hipLaunchKernelGGL(k_sweep_big, dim3((uint32_t)blocks), dim3(kBlock), 0,
+                           ctx->stream, A);
+      static const int sync_after_b = louvain_diag("MGX_LOUVAIN_SYNC_AFTER");
+      if (sync_after_b) {
+        hipError_t se = hipStreamSynchronize(ctx->stream);
+        hipError_t le = hipGetLastError();
+        if (se != hipSuccess || le != hipSuccess)
+          MGX_LTRACE("it=%lld BIG sync=%s last=%s", (long long)iters,
+                     hipGetErrorString(se), hipGetErrorString(le));
+      }
       MGX_LTRACE("it=%lld k_sweep_big launched", (long long)iters);
     }
     MGX_LDBG(ctx, "iter=%lld after-big", (long long)iters);
