@@ -246,6 +246,9 @@ struct SweepArgs {
   uint64_t pool_total;
   int inkernel_clear;        // experiment flag (MGX_LOUVAIN_INKERNEL_CLEAR)
   int64_t nv;                // level vertex count (community-id bound)
+  int diag_mask;             // storm-bisect arms (MGX_LOUVAIN_DIAG bitmask):
+                             // 1: no cupd atomics, 2: no e_xx add,
+                             // 4: targets forced to stay (sc)
   // Always-on guard accounting (4 u32 slots, device):
   //   [0] region-invariant violation bits (big-row pool geometry)
   //   [1] out-of-range max_index count (the suspected OOB vector of the
@@ -362,14 +365,15 @@ __global__ void __launch_bounds__(kBlock) k_sweep_small(SweepArgs A) {
       // swap protection (max(), utilityClusteringFunctions.cpp:305-307)
       if (A.cinfo_size[max_index] == 1 && A.cinfo_size[sc] == 1 && max_index > sc)
         max_index = sc;
+      if (A.diag_mask & 4) max_index = sc;
       A.target[row] = max_index;
-      if (max_index != sc) {
+      if (max_index != sc && !(A.diag_mask & 1)) {
         atomicAdd(&A.cupd_deg[max_index], A.vdeg[row]);
         atomicAdd(&A.cupd_size[max_index], 1);
         atomicAdd(&A.cupd_deg[sc], -A.vdeg[row]);
         atomicAdd(&A.cupd_size[sc], -1);
       }
-      atomicAdd(A.e_xx, own);
+      if (!(A.diag_mask & 2)) atomicAdd(A.e_xx, own);
     }
     wave_lds_fence();
   }
@@ -855,6 +859,10 @@ mgx_status louvain_level(mgx_context *ctx, const Level &L, double lower, double 
   A.pool_total = pool_total;
   A.inkernel_clear = 0;
   A.nv = nv;
+  {
+    const char *e = getenv("MGX_LOUVAIN_DIAG");
+    A.diag_mask = e ? atoi(e) : 0;
+  }
   DevBuf guard;
   {
     const char *e = getenv("MGX_LOUVAIN_INKERNEL_CLEAR");
