@@ -1061,6 +1061,8 @@ mgx_status renumber(mgx_context *ctx, int32_t *C, int64_t nv, int64_t *n_cluster
   unsigned long long n_active = 0;
   MGX_HIP_TRY(hipMemcpyAsync(&n_active, count.p, 8, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  MGX_LTRACE("renumber nv=%lld n_active=%llu", (long long)nv,
+             (unsigned long long)n_active);
   MGX_HIP_TRY(newid.alloc(ctx, nv * 4));
   hipLaunchKernelGGL(k_newid_from_sorted, dim3((uint32_t)grid_for((int64_t)n_active)),
                      dim3(kBlock), 0, ctx->stream, (int64_t)n_active,
@@ -1147,6 +1149,7 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   unsigned int n_unique = 0;
   MGX_HIP_TRY(hipMemcpyAsync(&n_unique, u_count.p, 4, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  MGX_LTRACE("coarsen ne2=%lld n_unique=%u", (long long)ne2, n_unique);
   // drop the sentinel group if present (it sorts last)
   int64_t n_pairs = n_unique;
   if (n_pairs > 0) {
@@ -1191,6 +1194,8 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   MGX_HIP_TRY(hipMemcpyAsync(out->row_ptr + n_clusters, &total, 4, hipMemcpyHostToDevice,
                              ctx->stream));
   out->ne2 = total;
+  MGX_LTRACE("coarsen n_pairs=%lld ncl=%lld total=%u", (long long)n_pairs,
+             (long long)n_clusters, total);
   MGX_HIP_TRY(hipMalloc(&out->col, (total > 0 ? total : 1) * 4));
   MGX_HIP_TRY(hipMalloc(&out->w, (total > 0 ? total : 1) * 8));
   // cursor = row_ptr copy (reuse counts buffer)
