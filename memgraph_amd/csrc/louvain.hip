@@ -1160,6 +1160,17 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
     if (last_key == sentinel) --n_pairs;
   }
 
+  if (louvain_trace() && n_pairs > 0) {
+    // probe: first/last unique keys + a counts sample after the passes
+    uint64_t k0 = 0, k1 = 0;
+    (void)hipMemcpyAsync(&k0, u_keys.as<uint64_t>(), 8, hipMemcpyDeviceToHost,
+                         ctx->stream);
+    (void)hipMemcpyAsync(&k1, u_keys.as<uint64_t>() + n_pairs - 1, 8,
+                         hipMemcpyDeviceToHost, ctx->stream);
+    (void)hipStreamSynchronize(ctx->stream);
+    MGX_LTRACE("coarsen ukeys[0]=(%d,%d) ukeys[last]=(%d,%d)", (int)(k0 >> 32),
+               (int)(uint32_t)k0, (int)(k1 >> 32), (int)(uint32_t)k1);
+  }
   // CSR counts + self presence.
   DevBuf counts, self_present;
   MGX_HIP_TRY(counts.alloc(ctx, n_clusters * 4));
@@ -1173,6 +1184,13 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   hipLaunchKernelGGL(k_add_missing_self, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock),
                      0, ctx->stream, n_clusters, self_present.as<uint32_t>(),
                      counts.as<uint32_t>());
+  if (louvain_trace()) {
+    uint32_t c0[4] = {0, 0, 0, 0};
+    (void)hipMemcpyAsync(c0, counts.p, 16, hipMemcpyDeviceToHost, ctx->stream);
+    hipError_t se = hipStreamSynchronize(ctx->stream);
+    MGX_LTRACE("coarsen counts[0..3]=%u,%u,%u,%u sync=%s", c0[0], c0[1], c0[2], c0[3],
+               hipGetErrorString(se));
+  }
 
   out->nv = n_clusters;
   MGX_HIP_TRY(hipMalloc(&out->row_ptr, (n_clusters + 1) * 4));
