@@ -559,13 +559,36 @@ __global__ void k_row_of_entry_big(int64_t n_big, const int32_t *rows,
   }
 }
 
+// diagnostic: count non-sentinel keys with components outside [0, ncl)
+__global__ void k_validate_keys(int64_t n, const uint64_t *keys, int64_t ncl,
+                                uint64_t sentinel, unsigned long long *bad) {
+  unsigned long long acc = 0;
+  for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n;
+       k += (int64_t)gridDim.x * blockDim.x) {
+    const uint64_t key = keys[k];
+    if (key == sentinel) continue;
+    const int64_t ci = (int64_t)(key >> 32);
+    const int64_t ct = (int64_t)(uint32_t)key;
+    if (ci < 0 || ci >= ncl || ct < 0 || ct >= ncl) ++acc;
+  }
+  if (acc) atomicAdd(bad, acc);
+}
+
 // count CSR rows for unique pairs (ci>=ct): row ci +1; if ct<ci also row ct +1.
-__global__ void k_pair_counts(int64_t n_pairs, const uint64_t *keys, uint32_t *counts,
-                              uint32_t *self_present) {
+__global__ void k_pair_counts(int64_t n_pairs, const uint64_t *keys, int64_t ncl,
+                              uint32_t *counts, uint32_t *self_present,
+                              unsigned long long *oob) {
   for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_pairs;
        k += (int64_t)gridDim.x * blockDim.x) {
     const int32_t ci = (int32_t)(keys[k] >> 32);
     const int32_t ct = (int32_t)(uint32_t)keys[k];
+    // Bounds: a corrupt pair id must be accounted, never become an OOB
+    // atomicAdd (the r02 storm fault vector — counts[1e9] faults or
+    // silently tramples a neighbouring allocation).
+    if (ci < 0 || ci >= ncl || ct < 0 || ct > ci) {
+      atomicAdd(oob, 1ull);
+      continue;
+    }
     atomicAdd(&counts[ci], 1u);
     if (ct < ci) atomicAdd(&counts[ct], 1u);
     else self_present[ci] = 1u;  // ct == ci
@@ -581,11 +604,12 @@ __global__ void k_add_missing_self(int64_t ncl, const uint32_t *self_present,
 }
 
 __global__ void k_pair_scatter(int64_t n_pairs, const uint64_t *keys, const double *vals,
-                               uint32_t *cursor, int32_t *col, double *w) {
+                               int64_t ncl, uint32_t *cursor, int32_t *col, double *w) {
   for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_pairs;
        k += (int64_t)gridDim.x * blockDim.x) {
     const int32_t ci = (int32_t)(keys[k] >> 32);
     const int32_t ct = (int32_t)(uint32_t)keys[k];
+    if (ci < 0 || ci >= ncl || ct < 0 || ct > ci) continue;  // accounted in counts
     uint32_t p = atomicAdd(&cursor[ci], 1u);
     col[p] = ct;
     w[p] = vals[k];
@@ -1119,6 +1143,19 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
                      sentinel);
   MGX_HIP_TRY(keys_sorted.alloc(ctx, ne2 * 8));
   MGX_HIP_TRY(vals_sorted.alloc(ctx, ne2 * 8));
+  DevBuf d_bad;
+  MGX_HIP_TRY(d_bad.alloc(ctx, 8));
+  MGX_HIP_TRY(hipMemsetAsync(d_bad.p, 0, 8, ctx->stream));
+  if (louvain_trace()) {
+    hipLaunchKernelGGL(k_validate_keys, dim3((uint32_t)grid_for(ne2)), dim3(kBlock), 0,
+                       ctx->stream, ne2, keys.as<uint64_t>(), n_clusters, sentinel,
+                       (unsigned long long *)d_bad.p);
+    unsigned long long bad = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&bad, d_bad.p, 8, hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    MGX_LTRACE("coarsen PAIRKEYS bad=%llu", bad);
+    MGX_HIP_TRY(hipMemsetAsync(d_bad.p, 0, 8, ctx->stream));
+  }
   size_t tmp_bytes = 0;
   auto err = rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys.as<uint64_t>(),
                                        keys_sorted.as<uint64_t>(), vals.as<double>(),
@@ -1150,6 +1187,17 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   MGX_HIP_TRY(hipMemcpyAsync(&n_unique, u_count.p, 4, hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   MGX_LTRACE("coarsen ne2=%lld n_unique=%u", (long long)ne2, n_unique);
+  if (louvain_trace() && n_unique > 0) {
+    hipLaunchKernelGGL(k_validate_keys, dim3((uint32_t)grid_for((int64_t)n_unique)),
+                       dim3(kBlock), 0, ctx->stream, (int64_t)n_unique,
+                       u_keys.as<uint64_t>(), n_clusters, sentinel,
+                       (unsigned long long *)d_bad.p);
+    unsigned long long bad = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&bad, d_bad.p, 8, hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    MGX_LTRACE("coarsen UKEYS bad=%llu", bad);
+    MGX_HIP_TRY(hipMemsetAsync(d_bad.p, 0, 8, ctx->stream));
+  }
   // drop the sentinel group if present (it sorts last)
   int64_t n_pairs = n_unique;
   if (n_pairs > 0) {
@@ -1179,8 +1227,26 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   MGX_HIP_TRY(hipMemsetAsync(self_present.p, 0, n_clusters * 4, ctx->stream));
   if (n_pairs > 0)
     hipLaunchKernelGGL(k_pair_counts, dim3((uint32_t)grid_for(n_pairs)), dim3(kBlock), 0,
-                       ctx->stream, n_pairs, u_keys.as<uint64_t>(), counts.as<uint32_t>(),
-                       self_present.as<uint32_t>());
+                       ctx->stream, n_pairs, u_keys.as<uint64_t>(), n_clusters,
+                       counts.as<uint32_t>(), self_present.as<uint32_t>(),
+                       (unsigned long long *)d_bad.p);
+  {
+    unsigned long long oob = 0;
+    MGX_HIP_TRY(hipMemcpyAsync(&oob, d_bad.p, 8, hipMemcpyDeviceToHost, ctx->stream));
+    MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+    if (oob) {
+      fprintf(stderr,
+              "[louvain] COARSEN GUARD: %llu out-of-range pair ids dropped "
+              "(ncl=%lld) — please report\n",
+              oob, (long long)n_clusters);
+      fflush(stderr);
+      const char *strict = getenv("MGX_LOUVAIN_STRICT");
+      if (strict && atoi(strict)) {
+        mgx_set_error("louvain coarsen guard fired (%llu bad pair ids)", oob);
+        return MGX_ERR_HIP;
+      }
+    }
+  }
   hipLaunchKernelGGL(k_add_missing_self, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock),
                      0, ctx->stream, n_clusters, self_present.as<uint32_t>(),
                      counts.as<uint32_t>());
@@ -1222,7 +1288,7 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   if (n_pairs > 0)
     hipLaunchKernelGGL(k_pair_scatter, dim3((uint32_t)grid_for(n_pairs)), dim3(kBlock), 0,
                        ctx->stream, n_pairs, u_keys.as<uint64_t>(), u_vals.as<double>(),
-                       counts.as<uint32_t>(), out->col, out->w);
+                       n_clusters, counts.as<uint32_t>(), out->col, out->w);
   hipLaunchKernelGGL(k_self_scatter, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock), 0,
                      ctx->stream, n_clusters, self_present.as<uint32_t>(),
                      counts.as<uint32_t>(), out->col, out->w);
