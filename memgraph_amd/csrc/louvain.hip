@@ -1274,6 +1274,16 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   MGX_HIP_TRY(hipMemcpyAsync(&last_cnt, counts.as<uint32_t>() + n_clusters - 1, 4,
                              hipMemcpyDeviceToHost, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
+  if (louvain_trace()) {
+    uint32_t rp[4] = {0, 0, 0, 0}, cs[4] = {0, 0, 0, 0};
+    (void)hipMemcpyAsync(rp, out->row_ptr, 16, hipMemcpyDeviceToHost, ctx->stream);
+    (void)hipMemcpyAsync(cs, counts.p, 16, hipMemcpyDeviceToHost, ctx->stream);
+    (void)hipStreamSynchronize(ctx->stream);
+    MGX_LTRACE("coarsen scan rp[0..3]=%u,%u,%u,%u cs[0..3]=%u,%u,%u,%u last_off=%u "
+               "last_cnt=%u",
+               rp[0], rp[1], rp[2], rp[3], cs[0], cs[1], cs[2], cs[3], last_off,
+               last_cnt);
+  }
   const uint32_t total = last_off + last_cnt;
   MGX_HIP_TRY(hipMemcpyAsync(out->row_ptr + n_clusters, &total, 4, hipMemcpyHostToDevice,
                              ctx->stream));
@@ -1285,15 +1295,18 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   // cursor = row_ptr copy (reuse counts buffer)
   MGX_HIP_TRY(hipMemcpyAsync(counts.p, out->row_ptr, n_clusters * 4,
                              hipMemcpyDeviceToDevice, ctx->stream));
+  MGX_LTRACE("coarsen pair_scatter");
   if (n_pairs > 0)
     hipLaunchKernelGGL(k_pair_scatter, dim3((uint32_t)grid_for(n_pairs)), dim3(kBlock), 0,
                        ctx->stream, n_pairs, u_keys.as<uint64_t>(), u_vals.as<double>(),
                        n_clusters, counts.as<uint32_t>(), out->col, out->w);
+  MGX_LTRACE("coarsen self_scatter");
   hipLaunchKernelGGL(k_self_scatter, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock), 0,
                      ctx->stream, n_clusters, self_present.as<uint32_t>(),
                      counts.as<uint32_t>(), out->col, out->w);
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   MGX_HIP_TRY(hipGetLastError());
+  MGX_LTRACE("coarsen done");
   return MGX_OK;
 }
 
