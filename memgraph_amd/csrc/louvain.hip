@@ -604,31 +604,38 @@ __global__ void k_add_missing_self(int64_t ncl, const uint32_t *self_present,
 }
 
 __global__ void k_pair_scatter(int64_t n_pairs, const uint64_t *keys, const double *vals,
-                               int64_t ncl, uint32_t *cursor, int32_t *col, double *w) {
+                               int64_t ncl, uint32_t total, uint32_t *cursor, int32_t *col,
+                               double *w) {
   for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < n_pairs;
        k += (int64_t)gridDim.x * blockDim.x) {
     const int32_t ci = (int32_t)(keys[k] >> 32);
     const int32_t ct = (int32_t)(uint32_t)keys[k];
     if (ci < 0 || ci >= ncl || ct < 0 || ct > ci) continue;  // accounted in counts
     uint32_t p = atomicAdd(&cursor[ci], 1u);
-    col[p] = ct;
-    w[p] = vals[k];
+    if (p < total) {
+      col[p] = ct;
+      w[p] = vals[k];
+    }
     if (ct < ci) {
       p = atomicAdd(&cursor[ct], 1u);
-      col[p] = ci;
-      w[p] = vals[k];
+      if (p < total) {
+        col[p] = ci;
+        w[p] = vals[k];
+      }
     }
   }
 }
 
-__global__ void k_self_scatter(int64_t ncl, const uint32_t *self_present, uint32_t *cursor,
-                               int32_t *col, double *w) {
+__global__ void k_self_scatter(int64_t ncl, const uint32_t *self_present, uint32_t total,
+                               uint32_t *cursor, int32_t *col, double *w) {
   for (int64_t c = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; c < ncl;
        c += (int64_t)gridDim.x * blockDim.x) {
     if (!self_present[c]) {
       uint32_t p = atomicAdd(&cursor[c], 1u);
-      col[p] = (int32_t)c;
-      w[p] = 0.0;  // zero-weight self loop (buildNextPhase.cpp cluPtrIn init)
+      if (p < total) {
+        col[p] = (int32_t)c;
+        w[p] = 0.0;  // zero-weight self loop (buildNextPhase.cpp cluPtrIn init)
+      }
     }
   }
 }
@@ -1260,6 +1267,25 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
 
   out->nv = n_clusters;
   MGX_HIP_TRY(hipMalloc(&out->row_ptr, (n_clusters + 1) * 4));
+  if (louvain_trace()) {
+    // Pre-scan probe: counts must still hold the pair histogram, and the
+    // freshly hipMalloc'd row_ptr must not alias the pool-allocated counts
+    // (the r02 corruption signature is an in-place exclusive scan -> zeros).
+    uint32_t c1[4] = {0, 0, 0, 0};
+    (void)hipMemcpyAsync(c1, counts.p, 16, hipMemcpyDeviceToHost, ctx->stream);
+    (void)hipStreamSynchronize(ctx->stream);
+    MGX_LTRACE("coarsen prescan counts[0..3]=%u,%u,%u,%u", c1[0], c1[1], c1[2], c1[3]);
+    const char *rp0 = (const char *)out->row_ptr;
+    const char *rp1 = rp0 + (size_t)(n_clusters + 1) * 4;
+    const char *cn0 = (const char *)counts.p, *cn1 = cn0 + (size_t)n_clusters * 4;
+    const char *ws0 = (const char *)ctx->workspace;
+    const char *ws1 = ws0 + ctx->workspace_bytes;
+    MGX_LTRACE("coarsen prescan row_ptr=[%p,%p) counts=[%p,%p) ws=[%p,%p)", rp0, rp1, cn0,
+               cn1, ws0, ws1);
+    if (rp0 < cn1 && cn0 < rp1) MGX_LTRACE("coarsen ALIAS row_ptr overlaps counts");
+    if (rp0 < ws1 && ws0 < rp1) MGX_LTRACE("coarsen ALIAS row_ptr overlaps workspace");
+    if (cn0 < ws1 && ws0 < cn1) MGX_LTRACE("coarsen ALIAS counts overlaps workspace");
+  }
   // exclusive scan counts -> row_ptr
   err = rocprim::exclusive_scan(nullptr, tmp_bytes, counts.as<uint32_t>(), out->row_ptr,
                                 0u, n_clusters, rocprim::plus<uint32_t>(), ctx->stream);
@@ -1290,8 +1316,56 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   out->ne2 = total;
   MGX_LTRACE("coarsen n_pairs=%lld ncl=%lld total=%u", (long long)n_pairs,
              (long long)n_clusters, total);
+  // Consistency gate: total = 2*offdiag + ncl >= max(n_pairs - dropped, ncl).
+  // A smaller total means the counts/row_ptr memory was trampled between the
+  // passes (the r02 storm signature: plain-hipMalloc row_ptr aliasing the
+  // pool-allocated counts block turned the exclusive scan in-place -> all
+  // zeros); scattering against it would write OOB, so fail cleanly instead.
+  if ((int64_t)total < n_clusters) {
+    mgx_set_error(
+        "louvain coarsen inconsistent (total=%u < ncl=%lld, n_pairs=%lld) — "
+        "device memory corruption detected",
+        total, (long long)n_clusters, (long long)n_pairs);
+    return MGX_ERR_HIP;
+  }
   MGX_HIP_TRY(hipMalloc(&out->col, (total > 0 ? total : 1) * 4));
   MGX_HIP_TRY(hipMalloc(&out->w, (total > 0 ? total : 1) * 8));
+  if (louvain_trace()) {
+    // Allocator-overlap probe: the async-pool DevBufs vs the plain hipMalloc
+    // level buffers must be disjoint VA ranges.
+    struct Rng {
+      const char *name;
+      const void *p;
+      size_t bytes;
+    } rngs[] = {
+        {"counts", counts.p, (size_t)n_clusters * 4},
+        {"self_present", self_present.p, (size_t)n_clusters * 4},
+        {"row_of_entry", row_of_entry.p, (size_t)ne2 * 4},
+        {"keys", keys.p, (size_t)ne2 * 8},
+        {"vals", vals.p, (size_t)ne2 * 8},
+        {"keys_sorted", keys_sorted.p, (size_t)ne2 * 8},
+        {"vals_sorted", vals_sorted.p, (size_t)ne2 * 8},
+        {"u_keys", u_keys.p, (size_t)ne2 * 8},
+        {"u_vals", u_vals.p, (size_t)ne2 * 8},
+        {"workspace", ctx->workspace, ctx->workspace_bytes},
+        {"in.row_ptr", in.row_ptr, (size_t)(in.nv + 1) * 4},
+        {"in.col", in.col, (size_t)ne2 * 4},
+        {"in.w", in.w, (size_t)ne2 * 8},
+        {"row_ptr", out->row_ptr, (size_t)(n_clusters + 1) * 4},
+        {"col", out->col, (size_t)(total > 0 ? total : 1) * 4},
+        {"w", out->w, (size_t)(total > 0 ? total : 1) * 8},
+    };
+    for (auto &r : rngs)
+      MGX_LTRACE("coarsen buf %-12s [%p, %p)", r.name, r.p,
+                 (const void *)((const char *)r.p + r.bytes));
+    for (size_t a = 0; a < sizeof(rngs) / sizeof(rngs[0]); ++a)
+      for (size_t b = a + 1; b < sizeof(rngs) / sizeof(rngs[0]); ++b) {
+        const char *a0 = (const char *)rngs[a].p, *a1 = a0 + rngs[a].bytes;
+        const char *b0 = (const char *)rngs[b].p, *b1 = b0 + rngs[b].bytes;
+        if (a0 < b1 && b0 < a1)
+          MGX_LTRACE("coarsen ALIAS %s overlaps %s", rngs[a].name, rngs[b].name);
+      }
+  }
   // cursor = row_ptr copy (reuse counts buffer)
   MGX_HIP_TRY(hipMemcpyAsync(counts.p, out->row_ptr, n_clusters * 4,
                              hipMemcpyDeviceToDevice, ctx->stream));
@@ -1299,10 +1373,10 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   if (n_pairs > 0)
     hipLaunchKernelGGL(k_pair_scatter, dim3((uint32_t)grid_for(n_pairs)), dim3(kBlock), 0,
                        ctx->stream, n_pairs, u_keys.as<uint64_t>(), u_vals.as<double>(),
-                       n_clusters, counts.as<uint32_t>(), out->col, out->w);
+                       n_clusters, total, counts.as<uint32_t>(), out->col, out->w);
   MGX_LTRACE("coarsen self_scatter");
   hipLaunchKernelGGL(k_self_scatter, dim3((uint32_t)grid_for(n_clusters)), dim3(kBlock), 0,
-                     ctx->stream, n_clusters, self_present.as<uint32_t>(),
+                     ctx->stream, n_clusters, self_present.as<uint32_t>(), total,
                      counts.as<uint32_t>(), out->col, out->w);
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   MGX_HIP_TRY(hipGetLastError());
