@@ -728,17 +728,17 @@ __global__ void k_check_marks(int64_t nv, const uint32_t *row_ptr, const uint32_
 // ---------- host orchestration -------------------------------------------
 
 struct DevBuf {
-  // Stream-ordered allocation from the cached mempool: per-level multi-GB
-  // hipMalloc/hipFree pairs cost ~2 s of host time per Louvain call at
-  // RMAT-24 (measured); the pool pays it once per process.
+  // Allocation from the context's caching free list (mgx_internal.h): the
+  // per-level multi-GB hipMalloc/hipFree pairs cost ~2 s of host time per
+  // Louvain call at RMAT-24 (measured); the cache pays it once per process.
   mgx_context *ctx = nullptr;
   void *p = nullptr;
   ~DevBuf() {
-    if (p) (void)hipFreeAsync(p, ctx->stream);
+    if (p) (void)ctx->free_async(p);
   }
   hipError_t alloc(mgx_context *c, size_t bytes) {
     ctx = c;
-    return hipMallocAsync(&p, bytes ? bytes : 1, c->stream);
+    return c->alloc_async(&p, bytes) == MGX_OK ? hipSuccess : hipErrorOutOfMemory;
   }
   template <typename T>
   T *as() {

@@ -41,48 +41,103 @@ extern "C" int mgx_device_count(void) {
 }
 
 namespace {
-bool no_async_pool() {
+// MGX_NO_ASYNC_POOL=1 bypasses the caching free list (one hipMalloc/hipFree
+// per request) — kept for allocator-bug bisection; the env name predates the
+// switch from hipMallocAsync to the in-house cache.
+bool no_alloc_cache() {
   static const bool v = [] {
     const char *e = getenv("MGX_NO_ASYNC_POOL");
     return e && atoi(e) != 0;
   }();
   return v;
 }
+
+// Bucket rounding: small requests to 512 B, >=2 MB to 2 MB granularity.
+// Louvain/graph-build reissue identical sizes level over level and call
+// over call, so rounded-exact reuse covers the hot pattern.
+size_t round_bucket(size_t bytes) {
+  if (bytes == 0) bytes = 1;
+  const size_t kSmall = 512, kBig = 2ull << 20;
+  if (bytes < kBig) return (bytes + kSmall - 1) / kSmall * kSmall;
+  return (bytes + kBig - 1) / kBig * kBig;
+}
 }  // namespace
 
 mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
-  if (no_async_pool()) {
+  if (no_alloc_cache()) {
     MGX_HIP_TRY(hipMalloc(p, bytes ? bytes : 1));
     return MGX_OK;
   }
-  MGX_HIP_TRY(hipMallocAsync(p, bytes ? bytes : 1, stream));
+  const size_t want = round_bucket(bytes);
+  {
+    std::lock_guard<std::mutex> lk(cache_mu);
+    auto it = cache_free.lower_bound(want);
+    // Accept a cached block up to 25% (+1 bucket) larger to limit waste.
+    if (it != cache_free.end() && it->first <= want + want / 4 + (2ull << 20)) {
+      *p = it->second;
+      cache_live[*p] = it->first;
+      cache_free_bytes -= it->first;
+      cache_free.erase(it);
+      return MGX_OK;
+    }
+  }
+  hipError_t e = hipMalloc(p, want);
+  if (e == hipErrorOutOfMemory) {
+    (void)hipGetLastError();
+    cache_trim();  // return every cached free block, then retry once
+    e = hipMalloc(p, want);
+  }
+  if (e != hipSuccess) {
+    (void)hipGetLastError();
+    mgx_set_error("hipMalloc(%zu) failed: %s", want, hipGetErrorString(e));
+    return e == hipErrorOutOfMemory ? MGX_ERR_OUT_OF_MEMORY : MGX_ERR_HIP;
+  }
+  std::lock_guard<std::mutex> lk(cache_mu);
+  cache_live[*p] = want;
   return MGX_OK;
 }
 
 mgx_status mgx_context::free_async(void *p) {
   if (!p) return MGX_OK;
-  if (no_async_pool()) {
+  if (no_alloc_cache()) {
     MGX_HIP_TRY(hipFree(p));
     return MGX_OK;
   }
-  MGX_HIP_TRY(hipFreeAsync(p, stream));
+  std::lock_guard<std::mutex> lk(cache_mu);
+  auto it = cache_live.find(p);
+  if (it == cache_live.end()) {
+    // Not cache-owned (pre-cache allocation): plain free.
+    MGX_HIP_TRY(hipFree(p));
+    return MGX_OK;
+  }
+  cache_free.insert({it->second, p});
+  cache_free_bytes += it->second;
+  cache_live.erase(it);
   return MGX_OK;
+}
+
+void mgx_context::cache_trim() {
+  // The blocks on the free list may still be referenced by queued stream
+  // work; drain before returning their pages.
+  (void)hipStreamSynchronize(stream);
+  std::lock_guard<std::mutex> lk(cache_mu);
+  for (auto &e : cache_free) (void)hipFree(e.second);
+  cache_free.clear();
+  cache_free_bytes = 0;
 }
 
 mgx_status mgx_context::reserve(size_t bytes, void **out) {
   if (bytes > workspace_bytes) {
-    // Drain queued users of the old workspace before freeing it. hipFree
-    // synchronizes the device on ROCm, but that contract is not worth
-    // betting stream-ordered correctness on; growth is rare so the sync
-    // costs nothing.
+    // Drain queued users of the old workspace before recycling it; growth
+    // is rare so the sync costs nothing.
     if (workspace) {
       MGX_HIP_TRY(hipStreamSynchronize(stream));
-      MGX_HIP_TRY(hipFree(workspace));
+      MGX_TRY(free_async(workspace));
     }
     workspace = nullptr;
     workspace_bytes = 0;
     size_t want = bytes + bytes / 2;
-    MGX_HIP_TRY(hipMalloc(&workspace, want));
+    MGX_TRY(alloc_async(&workspace, want));
     workspace_bytes = want;
   }
   *out = workspace;
@@ -108,12 +163,6 @@ extern "C" mgx_status mgx_init(int device, mgx_context **out) {
     mgx_set_error("hipStreamCreate failed");
     return MGX_ERR_HIP;
   }
-  // Cache freed stream-ordered allocations (see mgx_context::alloc_async).
-  hipMemPool_t pool = nullptr;
-  if (hipDeviceGetDefaultMemPool(&pool, device) == hipSuccess && pool) {
-    uint64_t threshold = UINT64_MAX;
-    (void)hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &threshold);
-  }
   *out = ctx;
   return MGX_OK;
 }
@@ -121,7 +170,9 @@ extern "C" mgx_status mgx_init(int device, mgx_context **out) {
 extern "C" mgx_status mgx_destroy(mgx_context *ctx) {
   if (!ctx) return MGX_OK;
   if (ctx->comm) (void)mgx_comm_destroy(ctx);
-  if (ctx->workspace) (void)hipFree(ctx->workspace);
+  if (ctx->workspace) (void)ctx->free_async(ctx->workspace);
+  ctx->cache_trim();
+  for (auto &e : ctx->cache_live) (void)hipFree(e.first);  // leaked by callers
   if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
   delete ctx;
   return MGX_OK;

@@ -6,7 +6,10 @@
 
 #include <cstdint>
 #include <cstdio>
+#include <map>
+#include <mutex>
 #include <string>
+#include <unordered_map>
 #include <vector>
 
 #include "../../include/mgx_analytics.h"
@@ -37,12 +40,26 @@ struct mgx_comm_state;  // comm.cpp (RCCL)
 struct mgx_context {
   int device = -1;
   hipStream_t stream = nullptr;
-  // Stream-ordered allocations for large transient buffers (sort keys, COO
-  // staging): the default mempool with an unbounded release threshold keeps
-  // freed blocks cached, so repeated multi-GB hipMalloc page-table setup
-  // (measured up to ~1 s per RMAT-26 build) is paid once per process.
+  // Caching allocator for large transient buffers (sort keys, COO staging):
+  // a size-bucketed free list over plain hipMalloc, so repeated multi-GB
+  // allocation page-table setup (measured up to ~1 s per RMAT-26 build) is
+  // paid once per process. Reuse is safe without syncs because every
+  // consumer enqueues on ctx->stream: the previous owner's work precedes
+  // the next owner's in stream order.
+  //
+  // Deliberately NOT hipMallocAsync: on the MI355X box's ROCm 7.0.x runtime
+  // a plain hipMalloc issued while the stream-ordered pool holds
+  // pending-free-then-reallocated blocks reclaims (and zeroes) the live
+  // blocks' pages — measured in the r02 Louvain storm (counts[] nonzero,
+  // then hipMalloc(row_ptr), then counts[] all-zero at disjoint VA; see
+  // profiles/r02_summary.md). One allocator family avoids the bug class.
+  std::mutex cache_mu;
+  std::unordered_map<void *, size_t> cache_live;  // ptr -> rounded bytes
+  std::multimap<size_t, void *> cache_free;       // rounded bytes -> ptr
+  size_t cache_free_bytes = 0;
   mgx_status alloc_async(void **p, size_t bytes);
   mgx_status free_async(void *p);
+  void cache_trim();  // hipFree every cached free block
   // Grow-only device workspace for rocPRIM temp storage etc.
   void *workspace = nullptr;
   size_t workspace_bytes = 0;
