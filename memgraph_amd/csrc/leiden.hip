@@ -324,6 +324,7 @@ struct RefineArgs {
   double *ext_w;
   double *edge_w;
   int32_t *nb_comms;
+  int32_t *nb_slot;  // rc -> 1+index into nb_comms, 0 = absent (O(1) lookup)
   double *prob;
   uint32_t *rsize;  // refined community local sizes
   double gamma;
@@ -342,11 +343,13 @@ __global__ void k_refine(RefineArgs A) {
     double *ext_w = A.ext_w + m0;
     double *edge_w = A.edge_w + m0;
     int32_t *nbc = A.nb_comms + m0;
+    int32_t *nslot = A.nb_slot + m0;
     double *prob = A.prob + m0;
     uint32_t *rsize = A.rsize + m0;
     for (uint32_t i = 0; i < k; ++i) {
       ext_w[i] = 0.0;
       edge_w[i] = 0.0;
+      nslot[i] = 0;
       rsize[i] = 1;
       A.rcomm_loc[A.members[m0 + i]] = (int32_t)i;
     }
@@ -374,14 +377,13 @@ __global__ void k_refine(RefineArgs A) {
         const int32_t u = A.col[j];
         if (A.comm[u] != (int32_t)c) continue;
         const int32_t rc = A.rcomm_loc[u];
-        if (edge_w[rc] == 0.0) {
-          bool seen = false;
-          for (int32_t q = 0; q < nn; ++q)
-            if (nbc[q] == rc) {
-              seen = true;
-              break;
-            }
-          if (!seen) nbc[nn++] = rc;
+        // First-encounter append, O(1) via the slot table (semantically
+        // identical to a linear membership scan of nbc[0..nn), which is
+        // O(deg^2) for hub members and hung RMAT-scale runs).
+        if (edge_w[rc] == 0.0 && nslot[rc] == 0) {
+          nbc[nn] = rc;
+          nslot[rc] = nn + 1;
+          ++nn;
         }
         edge_w[rc] += A.w[j];
       }
@@ -409,6 +411,7 @@ __global__ void k_refine(RefineArgs A) {
         p = total;
         prob[q] = p;  // cumulative snapshot per neighbour slot (:264)
         edge_w[rc] = 0.0;
+        nslot[rc] = 0;
       }
       if (nn > 0) {
         // the reference's always-true branch (:268-276): draw in [0,total)
@@ -710,7 +713,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       int32_t *d_members = nullptr, *d_locidx = nullptr, *d_rcomm = nullptr;
       uint32_t *d_memoff = nullptr, *d_rsize = nullptr;
       double *d_extw = nullptr, *d_edgew = nullptr, *d_prob = nullptr;
-      int32_t *d_nbc = nullptr;
+      int32_t *d_nbc = nullptr, *d_cslot = nullptr;
       MGX_HIP_TRY(hipMalloc(&d_members, nv * 4));
       MGX_HIP_TRY(hipMalloc(&d_locidx, nv * 4));
       MGX_HIP_TRY(hipMalloc(&d_rcomm, nv * 4));
@@ -720,6 +723,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       MGX_HIP_TRY(hipMalloc(&d_edgew, nv * 8));
       MGX_HIP_TRY(hipMalloc(&d_prob, nv * 8));
       MGX_HIP_TRY(hipMalloc(&d_nbc, nv * 4));
+      MGX_HIP_TRY(hipMalloc(&d_cslot, nv * 4));
       MGX_HIP_TRY(hipMemcpyAsync(d_members, members.data(), nv * 4,
                                  hipMemcpyHostToDevice, ctx->stream));
       MGX_HIP_TRY(hipMemcpyAsync(d_memoff, mem_off.data(), (nv + 1) * 4,
@@ -742,6 +746,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       R.ext_w = d_extw;
       R.edge_w = d_edgew;
       R.nb_comms = d_nbc;
+      R.nb_slot = d_cslot;
       R.prob = d_prob;
       R.rsize = d_rsize;
       R.gamma = gamma_n;
@@ -765,6 +770,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       (void)hipFree(d_edgew);
       (void)hipFree(d_prob);
       (void)hipFree(d_nbc);
+      (void)hipFree(d_cslot);
       (void)hipFree(d_rcomm);
 
       // global refined community id per node: the member slot of its local
