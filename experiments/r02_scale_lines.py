@@ -40,12 +40,23 @@ if "leiden" in algos:
     bench(f"leiden RMAT-{SCALE}w", BUILD_SYM_CSR | BUILD_WEIGHTED, run_leiden)
 
 if "lrt" in algos:
+    # LabelRankT seeds deg+1 labels per node (the reference's init), so the
+    # first propagate is Theta(sum deg^2): infeasible on power-law RMAT for
+    # any implementation. Measure on a bounded-degree uniform graph.
     def run_lrt(g):
         comm = n.lrt_set(ctx, g, ids, directed=False, weighted=True)
         k = len(np.unique(comm))
         n.lrt_reset(ctx)
         return f"{k} communities"
-    bench(f"labelrankt RMAT-{SCALE}w", BUILD_SYM_CSR | BUILD_WEIGHTED, run_lrt)
+    g = n.graph_uniform(ctx, V, E, seed=42, flags=BUILD_SYM_CSR | BUILD_WEIGHTED)
+    print(f"[labelrankt UNIFORM-{SCALE}w] graph built ({n.graph_build_ms(g):.0f} ms)",
+          flush=True)
+    import time as _t
+    _t0 = _t.perf_counter()
+    out = run_lrt(g)
+    print(f"[labelrankt UNIFORM-{SCALE}w] {_t.perf_counter()-_t0:.2f} s :: {out}",
+          flush=True)
+    n.graph_destroy(ctx, g)
 
 if "pronline" in algos:
     def run_pronline(g):

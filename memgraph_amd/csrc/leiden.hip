@@ -40,8 +40,8 @@
 // under 256 neighbours, a global pool region per hub row.
 
 #include <cstring>
-#include <set>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 #include <rocprim/rocprim.hpp>
@@ -836,15 +836,17 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
           MGX_HIP_TRY(hipMemcpyAsync(w_l.data(), L.w, L.ne2 * 8, hipMemcpyDeviceToHost,
                                      ctx->stream));
           MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-          std::set<std::pair<int32_t, int32_t>> seen;
+          std::unordered_set<uint64_t> seen;
+          seen.reserve((size_t)(L.ne2 / 2 + 16));
           for (int64_t v = 0; v < nv; ++v) {
             const int32_t cu = (int32_t)parent[v];
             for (uint32_t j = rp_l[v]; j < rp_l[v + 1]; ++j) {
               const int32_t cv2 = (int32_t)parent[col_l[j]];
               if (cu == cv2) continue;
-              auto key = cu < cv2 ? std::make_pair(cu, cv2) : std::make_pair(cv2, cu);
-              if (seen.count(key)) continue;
-              seen.insert(key);
+              const uint64_t key = cu < cv2
+                                       ? ((uint64_t)(uint32_t)cu << 32) | (uint32_t)cv2
+                                       : ((uint64_t)(uint32_t)cv2 << 32) | (uint32_t)cu;
+              if (!seen.insert(key).second) continue;
               adj[cu].emplace_back(cv2, w_l[j]);
               adj[cv2].emplace_back(cu, w_l[j]);
               nw_new[cu] += w_l[j];

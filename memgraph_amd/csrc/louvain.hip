@@ -1266,6 +1266,7 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   }
 
   out->nv = n_clusters;
+  ctx->ensure_margin((size_t)(n_clusters + 1) * 4);
   MGX_HIP_TRY(hipMalloc(&out->row_ptr, (n_clusters + 1) * 4));
   if (louvain_trace()) {
     // Pre-scan probe: counts must still hold the pair histogram, and the
@@ -1328,6 +1329,7 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
         total, (long long)n_clusters, (long long)n_pairs);
     return MGX_ERR_HIP;
   }
+  ctx->ensure_margin((size_t)(total > 0 ? total : 1) * 12);
   MGX_HIP_TRY(hipMalloc(&out->col, (total > 0 ? total : 1) * 4));
   MGX_HIP_TRY(hipMalloc(&out->w, (total > 0 ? total : 1) * 8));
   if (louvain_trace()) {

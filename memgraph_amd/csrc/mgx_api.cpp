@@ -81,6 +81,7 @@ mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
       return MGX_OK;
     }
   }
+  ensure_margin(want);
   hipError_t e = hipMalloc(p, want);
   if (e == hipErrorOutOfMemory) {
     (void)hipGetLastError();
@@ -114,6 +115,18 @@ mgx_status mgx_context::free_async(void *p) {
   cache_free_bytes += it->second;
   cache_live.erase(it);
   return MGX_OK;
+}
+
+void mgx_context::ensure_margin(size_t want) {
+  // Near device-memory exhaustion the ROCm 7.0.x runtime can hand out
+  // bogus mappings instead of failing (overlapping VA / read-only pages —
+  // measured at RMAT-26 Louvain, see profiles/r02_summary.md). Keep a
+  // safety margin: release every cached free block before a malloc that
+  // would leave less than ~8 GB free.
+  if (cache_free_bytes == 0) return;
+  size_t free_b = 0, total_b = 0;
+  if (hipMemGetInfo(&free_b, &total_b) == hipSuccess && free_b < want + (8ull << 30))
+    cache_trim();
 }
 
 void mgx_context::cache_trim() {

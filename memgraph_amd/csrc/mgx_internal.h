@@ -60,6 +60,9 @@ struct mgx_context {
   mgx_status alloc_async(void **p, size_t bytes);
   mgx_status free_async(void *p);
   void cache_trim();  // hipFree every cached free block
+  // Release the cache if a coming plain hipMalloc of `want` bytes would
+  // leave the device under the safety margin (see alloc_async).
+  void ensure_margin(size_t want);
   // Grow-only device workspace for rocPRIM temp storage etc.
   void *workspace = nullptr;
   size_t workspace_bytes = 0;
