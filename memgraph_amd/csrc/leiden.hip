@@ -51,6 +51,23 @@
 
 namespace {
 
+// MGX_LEIDEN_TRACE=1: per-level phase markers on stderr (diagnostics only).
+inline bool leiden_trace() {
+  static const bool v = [] {
+    const char *e = getenv("MGX_LEIDEN_TRACE");
+    return e && atoi(e) != 0;
+  }();
+  return v;
+}
+#define MGX_LEIDEN_LOG(...)                \
+  do {                                     \
+    if (leiden_trace()) {                  \
+      fprintf(stderr, "[leiden] " __VA_ARGS__); \
+      fprintf(stderr, "\n");               \
+      fflush(stderr);                      \
+    }                                      \
+  } while (0)
+
 constexpr int kBlock = 256;
 constexpr uint32_t kSmallRowDeg = 256;
 constexpr int kLdsCap = 512;
@@ -561,6 +578,8 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
   while (!done && st == MGX_OK) {
     ++iters;
     const int64_t nv = L.nv;
+    MGX_LEIDEN_LOG("level %lld nv=%lld ne2=%lld", (long long)level, (long long)nv,
+                   (long long)L.ne2);
     // ---- MoveNodesFast (red-black Jacobi to fixed point) ----------------
     // row classification for the sweep kernels
     std::vector<uint32_t> rp_l(nv + 1);
@@ -636,7 +655,9 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     M.pool_keys = d_pkeys;
     M.pool_vals = d_pvals;
     M.pool_off = d_poff;
+    int sweeps_done = 0;
     for (int sweep = 0; sweep < 200; ++sweep) {
+      ++sweeps_done;
       unsigned long long moved_total = 0;
       for (int parity = 0; parity < 2; ++parity) {
         M.parity = parity;
@@ -667,7 +688,10 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
         moved_total += moved;
       }
       if (moved_total == 0) break;
+      if (leiden_trace() && (sweep & 15) == 0)
+        MGX_LEIDEN_LOG("  sweep %d moved=%llu", sweep, moved_total);
     }
+    MGX_LEIDEN_LOG("  moves done (%d sweeps)", sweeps_done);
 
     // singleton / empties accounting
     unsigned long long nonsingleton = 0, nonempty = 0;
@@ -754,6 +778,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       R.resolution = resolution;
       R.seed = mgx_hash64(ms, (uint64_t)level + 1);
       R.n_merged = d_cnt;
+      MGX_LEIDEN_LOG("  refine launch n_comm=%lld", (long long)nv);
       hipLaunchKernelGGL(k_refine, dim3((uint32_t)grid_for(nv)), dim3(kBlock), 0,
                          ctx->stream, R);
       unsigned long long merged = 0;
@@ -802,6 +827,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       if (!done) {
         const std::vector<int32_t> &agg_refined = use_move_partition ? comm_h : refined;
         const std::vector<int32_t> &agg_original = comm_h;
+        MGX_LEIDEN_LOG("  aggregate start");
         // ---- aggregate (host; level sizes shrink fast) ------------------
         // compact refined community ids
         std::unordered_map<int32_t, int32_t> remap;
