@@ -318,7 +318,7 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
   while (cap < need) cap *= 2;
   auto grow = [&](double **p, double init) -> mgx_status {
     double *np = nullptr;
-    MGX_HIP_TRY(hipMalloc(&np, cap * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&np, cap * 8));
     hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(cap)), dim3(kBlock), 0,
                        ctx->stream, cap, init, np);
     if (*p && g_k.slots_cap > 0) {
@@ -339,7 +339,7 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
   MGX_TRY(grow(&g_k.ur, 0.0));
   {
     uint8_t *na = nullptr;
-    MGX_HIP_TRY(hipMalloc(&na, cap));
+    MGX_HIP_TRY(mgx_hip_malloc(&na, cap));
     MGX_HIP_TRY(hipMemsetAsync(na, 0, cap, ctx->stream));
     if (g_k.d_active && g_k.slots_cap > 0) {
       MGX_HIP_TRY(hipMemcpyAsync(na, g_k.d_active, g_k.slots_cap,
@@ -355,8 +355,8 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
 
 mgx_status kon_add_iteration(mgx_context *ctx, double omega_init) {
   double *w = nullptr, *c = nullptr;
-  MGX_HIP_TRY(hipMalloc(&w, g_k.slots_cap * 8));
-  MGX_HIP_TRY(hipMalloc(&c, g_k.slots_cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&w, g_k.slots_cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&c, g_k.slots_cap * 8));
   hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(g_k.slots_cap)), dim3(kBlock), 0,
                      ctx->stream, g_k.slots_cap, omega_init, w);
   hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(g_k.slots_cap)), dim3(kBlock), 0,
@@ -400,7 +400,7 @@ mgx_status kon_build_maps(mgx_context *ctx, mgx_graph *g, const int64_t *dense_t
     }
   }
   if (all_known) *all_known = known;
-  MGX_HIP_TRY(hipMalloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m->dense2slot, m->d2s_h.data(), (V > 0 ? V : 1) * 4,
                              hipMemcpyHostToDevice, ctx->stream));
   // in-degree bins from the in-CSR (host-side row_ptr copy)
@@ -417,8 +417,8 @@ mgx_status kon_build_maps(mgx_context *ctx, mgx_graph *g, const int64_t *dense_t
     }
     m->n_small = (int64_t)sm.size();
     m->n_big = (int64_t)bg.size();
-    MGX_HIP_TRY(hipMalloc(&m->small_rows, (m->n_small > 0 ? m->n_small : 1) * 4));
-    MGX_HIP_TRY(hipMalloc(&m->big_rows, (m->n_big > 0 ? m->n_big : 1) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&m->small_rows, (m->n_small > 0 ? m->n_small : 1) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&m->big_rows, (m->n_big > 0 ? m->n_big : 1) * 4));
     // empty vectors have a null data(): copy only when non-empty
     if (m->n_small > 0)
       MGX_HIP_TRY(hipMemcpyAsync(m->small_rows, sm.data(), m->n_small * 4,
@@ -443,11 +443,11 @@ mgx_status kon_loop(mgx_context *ctx, mgx_graph *g, const KMaps &m, double gamma
   int32_t *d_act = nullptr, *d_act_sorted = nullptr;
   uint64_t *d_keys = nullptr, *d_keys_sorted = nullptr;
   uint32_t *d_flag = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_act, (n_active > 0 ? n_active : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&d_act_sorted, (n_active > 0 ? n_active : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&d_keys, (n_active > 0 ? n_active : 1) * 8));
-  MGX_HIP_TRY(hipMalloc(&d_keys_sorted, (n_active > 0 ? n_active : 1) * 8));
-  MGX_HIP_TRY(hipMalloc(&d_flag, 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_act, (n_active > 0 ? n_active : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_act_sorted, (n_active > 0 ? n_active : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_keys, (n_active > 0 ? n_active : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_keys_sorted, (n_active > 0 ? n_active : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_flag, 4));
   if (n_active > 0)
     MGX_HIP_TRY(hipMemcpyAsync(d_act, act.data(), n_active * 4, hipMemcpyHostToDevice,
                                ctx->stream));
@@ -611,7 +611,7 @@ extern "C" mgx_status mgx_konline_get(mgx_context *ctx, const int64_t *dense_to_
   if (!ok || !out || V == 0) return MGX_OK;
   KMaps m;
   m.V = V;
-  MGX_HIP_TRY(hipMalloc(&m.dense2slot, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m.dense2slot, V * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m.dense2slot, d2s.data(), V * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   return kon_output(ctx, m, out);
@@ -697,8 +697,8 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
   }
   uint64_t *d_pairs = nullptr;
   uint32_t *d_cnt = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_pairs, (pair_keys.empty() ? 1 : pair_keys.size()) * 8));
-  MGX_HIP_TRY(hipMalloc(&d_cnt, (pair_cnt.empty() ? 1 : pair_cnt.size()) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_pairs, (pair_keys.empty() ? 1 : pair_keys.size()) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_cnt, (pair_cnt.empty() ? 1 : pair_cnt.size()) * 4));
   if (!pair_keys.empty()) {
     MGX_HIP_TRY(hipMemcpyAsync(d_pairs, pair_keys.data(), pair_keys.size() * 8,
                                hipMemcpyHostToDevice, ctx->stream));
@@ -723,7 +723,7 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
     del_vs.push_back(v->second);
   }
   auto upload_i32 = [&](const std::vector<int32_t> &v, int32_t **d) -> mgx_status {
-    MGX_HIP_TRY(hipMalloc(d, (v.empty() ? 1 : v.size()) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(d, (v.empty() ? 1 : v.size()) * 4));
     if (!v.empty())
       MGX_HIP_TRY(hipMemcpyAsync(*d, v.data(), v.size() * 4, hipMemcpyHostToDevice,
                                  ctx->stream));
@@ -744,7 +744,7 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
       if (p) (void)hipFree(p);
   };
   for (int64_t i = 0; i <= g_k.iteration; ++i) {
-    if (hipMalloc(&new_omega[i], g_k.slots_cap * 8) != hipSuccess) {
+    if (mgx_hip_malloc(&new_omega[i], g_k.slots_cap * 8) != hipSuccess) {
       cleanup_new();
       mgx_set_error("konline_update: out of memory for level omegas");
       return MGX_ERR_OUT_OF_MEMORY;
@@ -757,10 +757,10 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
   uint32_t *d_updated = nullptr;
   int32_t *d_front = nullptr, *d_next = nullptr;
   unsigned long long *d_nn = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_updated, V * 4));
-  MGX_HIP_TRY(hipMalloc(&d_front, V * 4));
-  MGX_HIP_TRY(hipMalloc(&d_next, V * 4));
-  MGX_HIP_TRY(hipMalloc(&d_nn, 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_updated, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_front, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_next, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_nn, 8));
   std::vector<uint32_t> upd_h(V, 0);
   std::vector<int32_t> front_h;
   for (auto mg : updated_mg) {

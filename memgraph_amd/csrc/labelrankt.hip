@@ -414,10 +414,10 @@ mgx_status lrt_grow_slots(mgx_context *ctx, int64_t need) {
   uint64_t *noff = nullptr;
   uint32_t *nlen = nullptr, *ntu = nullptr;
   double *nsw = nullptr;
-  MGX_HIP_TRY(hipMalloc(&noff, cap * 8));
-  MGX_HIP_TRY(hipMalloc(&nlen, cap * 4));
-  MGX_HIP_TRY(hipMalloc(&ntu, cap * 4));
-  MGX_HIP_TRY(hipMalloc(&nsw, cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&noff, cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&nlen, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&ntu, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nsw, cap * 8));
   MGX_HIP_TRY(hipMemsetAsync(nlen, 0, cap * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(ntu, 0, cap * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(noff, 0, cap * 8, ctx->stream));
@@ -479,7 +479,7 @@ mgx_status lrt_build_maps(mgx_context *ctx, int64_t V, const int64_t *dense_to_m
   m->d2s_h.assign(V > 0 ? V : 1, -1);
   for (int64_t v = 0; v < V; ++v) m->d2s_h[v] = lrt_slot(dense_to_mg[v]);
   MGX_TRY(lrt_grow_slots(ctx, (int64_t)g_l.slot2mg.size()));
-  MGX_HIP_TRY(hipMalloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m->dense2slot, m->d2s_h.data(), (V > 0 ? V : 1) * 4,
                              hipMemcpyHostToDevice, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
@@ -497,8 +497,8 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
   if (n == 0) return MGX_OK;
   int32_t *d_nodes = nullptr;
   uint32_t *d_counts = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_nodes, n * 4));
-  MGX_HIP_TRY(hipMalloc(&d_counts, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_nodes, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_counts, n * 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_nodes, nodes_dense.data(), n * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   hipLaunchKernelGGL(k_set_count, dim3((uint32_t)grid_for(n)), dim3(kBlock), 0,
@@ -513,8 +513,8 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
     while (cap < need) cap *= 2;
     int32_t *nl = nullptr;
     double *np = nullptr;
-    MGX_HIP_TRY(hipMalloc(&nl, cap * 4));
-    MGX_HIP_TRY(hipMalloc(&np, cap * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&nl, cap * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&np, cap * 8));
     if (g_l.pool_used > 0) {
       MGX_HIP_TRY(hipMemcpyAsync(nl, g_l.lab, g_l.pool_used * 4,
                                  hipMemcpyDeviceToDevice, ctx->stream));
@@ -530,7 +530,7 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
   }
   for (auto &o : seg_off) o += (uint64_t)g_l.pool_used;
   uint64_t *d_seg = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_seg, n * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_seg, n * 8));
   MGX_HIP_TRY(hipMemcpyAsync(d_seg, seg_off.data(), n * 8, hipMemcpyHostToDevice,
                              ctx->stream));
   SetArgs A;
@@ -572,9 +572,9 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   int32_t *d_nodes = nullptr;
   uint8_t *d_sel = nullptr;
   uint32_t *d_bound = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_nodes, n * 4));
-  MGX_HIP_TRY(hipMalloc(&d_sel, n));
-  MGX_HIP_TRY(hipMalloc(&d_bound, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_nodes, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_sel, n));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_bound, n * 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_nodes, candidates.data(), n * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   hipLaunchKernelGGL(k_distinct, dim3((uint32_t)grid_for(n)), dim3(kBlock), 0, ctx->stream,
@@ -603,10 +603,10 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   double *scr_p = nullptr;
   uint64_t *d_scr_off = nullptr;
   uint32_t *d_cnt = nullptr;
-  MGX_HIP_TRY(hipMalloc(&scr_lab, (scr_total > 0 ? scr_total : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&scr_p, (scr_total > 0 ? scr_total : 1) * 8));
-  MGX_HIP_TRY(hipMalloc(&d_scr_off, n * 8));
-  MGX_HIP_TRY(hipMalloc(&d_cnt, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&scr_lab, (scr_total > 0 ? scr_total : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&scr_p, (scr_total > 0 ? scr_total : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_scr_off, n * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_cnt, n * 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_scr_off, scr_off.data(), n * 8, hipMemcpyHostToDevice,
                              ctx->stream));
   PropArgs P;
@@ -677,14 +677,14 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   uint32_t *d_out_len = nullptr;
   unsigned long long *d_most = nullptr;
   uint32_t *d_any = nullptr;
-  MGX_HIP_TRY(hipMalloc(&nlab, (total > 0 ? total : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&np, (total > 0 ? total : 1) * 8));
-  MGX_HIP_TRY(hipMalloc(&d_new_off, n_slots * 8));
-  MGX_HIP_TRY(hipMalloc(&d_slot_sel, n_slots * 4));
-  MGX_HIP_TRY(hipMalloc(&d_out_off, n_slots * 8));
-  MGX_HIP_TRY(hipMalloc(&d_out_len, n_slots * 4));
-  MGX_HIP_TRY(hipMalloc(&d_most, 8));
-  MGX_HIP_TRY(hipMalloc(&d_any, 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nlab, (total > 0 ? total : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&np, (total > 0 ? total : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_new_off, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_slot_sel, n_slots * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_out_off, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_out_len, n_slots * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_most, 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_any, 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_new_off, new_off.data(), n_slots * 8,
                              hipMemcpyHostToDevice, ctx->stream));
   MGX_HIP_TRY(hipMemcpyAsync(d_slot_sel, slot_sel.data(), n_slots * 4,
@@ -749,8 +749,8 @@ mgx_status lrt_all_labels(mgx_context *ctx, const LrtMaps &m, int64_t *out_label
   const int64_t n_slots = (int64_t)g_l.slot2mg.size();
   if (n_slots == 0 || m.V == 0) return MGX_OK;
   int64_t *d_s2mg = nullptr, *d_raw = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_s2mg, n_slots * 8));
-  MGX_HIP_TRY(hipMalloc(&d_raw, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_s2mg, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_raw, n_slots * 8));
   MGX_HIP_TRY(hipMemcpyAsync(d_s2mg, g_l.slot2mg.data(), n_slots * 8,
                              hipMemcpyHostToDevice, ctx->stream));
   hipLaunchKernelGGL(k_argmax, dim3((uint32_t)grid_for(n_slots)), dim3(kBlock), 0,

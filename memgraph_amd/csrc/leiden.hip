@@ -539,10 +539,10 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
   Level L;
   L.nv = V;
   L.ne2 = (int64_t)col2.size();
-  MGX_HIP_TRY(hipMalloc(&L.row_ptr, (V + 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&L.col, (L.ne2 > 0 ? L.ne2 : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&L.w, (L.ne2 > 0 ? L.ne2 : 1) * 8));
-  MGX_HIP_TRY(hipMalloc(&L.node_w, V * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.row_ptr, (V + 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.col, (L.ne2 > 0 ? L.ne2 : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.w, (L.ne2 > 0 ? L.ne2 : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.node_w, V * 8));
   MGX_HIP_TRY(hipMemcpyAsync(L.row_ptr, rp2.data(), (V + 1) * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   if (L.ne2 > 0) {
@@ -568,8 +568,8 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
   // it; at level 0 it is singleton.
   int32_t *d_comm = nullptr;
   uint32_t *d_csize = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_comm, V * 4));
-  MGX_HIP_TRY(hipMalloc(&d_csize, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_comm, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_csize, V * 4));
   hipLaunchKernelGGL(k_iota32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
                      V, d_comm);
   hipLaunchKernelGGL(k_fillu32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
@@ -609,13 +609,13 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     int32_t *d_pkeys = nullptr;
     double *d_pvals = nullptr;
     unsigned long long *d_cnt = nullptr;
-    MGX_HIP_TRY(hipMalloc(&d_srows, (srows.empty() ? 1 : srows.size()) * 4));
-    MGX_HIP_TRY(hipMalloc(&d_brows, (brows.empty() ? 1 : brows.size()) * 4));
-    MGX_HIP_TRY(hipMalloc(&d_target, nv * 4));
-    MGX_HIP_TRY(hipMalloc(&d_poff, caps.size() * 8));
-    MGX_HIP_TRY(hipMalloc(&d_pkeys, (pool_total ? pool_total : 1) * 4));
-    MGX_HIP_TRY(hipMalloc(&d_pvals, (pool_total ? pool_total : 1) * 8));
-    MGX_HIP_TRY(hipMalloc(&d_cnt, 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_srows, (srows.empty() ? 1 : srows.size()) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_brows, (brows.empty() ? 1 : brows.size()) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_target, nv * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_poff, caps.size() * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_pkeys, (pool_total ? pool_total : 1) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_pvals, (pool_total ? pool_total : 1) * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_cnt, 8));
     if (!srows.empty())
       MGX_HIP_TRY(hipMemcpyAsync(d_srows, srows.data(), srows.size() * 4,
                                  hipMemcpyHostToDevice, ctx->stream));
@@ -628,7 +628,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     unsigned long long empties_before = 0;
     {
       unsigned long long *d_t = nullptr;
-      MGX_HIP_TRY(hipMalloc(&d_t, 16));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_t, 16));
       MGX_HIP_TRY(hipMemsetAsync(d_t, 0, 16, ctx->stream));
       hipLaunchKernelGGL(k_count_nonsingleton, dim3((uint32_t)grid_for(nv)), dim3(kBlock),
                          0, ctx->stream, nv, d_csize, d_t, d_t + 1);
@@ -697,7 +697,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     unsigned long long nonsingleton = 0, nonempty = 0;
     {
       unsigned long long *d_t = nullptr;
-      MGX_HIP_TRY(hipMalloc(&d_t, 16));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_t, 16));
       MGX_HIP_TRY(hipMemsetAsync(d_t, 0, 16, ctx->stream));
       hipLaunchKernelGGL(k_count_nonsingleton, dim3((uint32_t)grid_for(nv)), dim3(kBlock),
                          0, ctx->stream, nv, d_csize, d_t, d_t + 1);
@@ -738,16 +738,16 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       uint32_t *d_memoff = nullptr, *d_rsize = nullptr;
       double *d_extw = nullptr, *d_edgew = nullptr, *d_prob = nullptr;
       int32_t *d_nbc = nullptr, *d_cslot = nullptr;
-      MGX_HIP_TRY(hipMalloc(&d_members, nv * 4));
-      MGX_HIP_TRY(hipMalloc(&d_locidx, nv * 4));
-      MGX_HIP_TRY(hipMalloc(&d_rcomm, nv * 4));
-      MGX_HIP_TRY(hipMalloc(&d_memoff, (nv + 1) * 4));
-      MGX_HIP_TRY(hipMalloc(&d_rsize, nv * 4));
-      MGX_HIP_TRY(hipMalloc(&d_extw, nv * 8));
-      MGX_HIP_TRY(hipMalloc(&d_edgew, nv * 8));
-      MGX_HIP_TRY(hipMalloc(&d_prob, nv * 8));
-      MGX_HIP_TRY(hipMalloc(&d_nbc, nv * 4));
-      MGX_HIP_TRY(hipMalloc(&d_cslot, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_members, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_locidx, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_rcomm, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_memoff, (nv + 1) * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_rsize, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_extw, nv * 8));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_edgew, nv * 8));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_prob, nv * 8));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_nbc, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_cslot, nv * 4));
       MGX_HIP_TRY(hipMemcpyAsync(d_members, members.data(), nv * 4,
                                  hipMemcpyHostToDevice, ctx->stream));
       MGX_HIP_TRY(hipMemcpyAsync(d_memoff, mem_off.data(), (nv + 1) * 4,
@@ -914,10 +914,10 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
           rp_up[i2 + 1] = (uint32_t)col_up.size();
         }
         L.ne2 = (int64_t)col_up.size();
-        MGX_HIP_TRY(hipMalloc(&L.row_ptr, (n_new + 1) * 4));
-        MGX_HIP_TRY(hipMalloc(&L.col, (L.ne2 ? L.ne2 : 1) * 4));
-        MGX_HIP_TRY(hipMalloc(&L.w, (L.ne2 ? L.ne2 : 1) * 8));
-        MGX_HIP_TRY(hipMalloc(&L.node_w, n_new * 8));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.row_ptr, (n_new + 1) * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.col, (L.ne2 ? L.ne2 : 1) * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.w, (L.ne2 ? L.ne2 : 1) * 8));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.node_w, n_new * 8));
         MGX_HIP_TRY(hipMemcpyAsync(L.row_ptr, rp_up.data(), (n_new + 1) * 4,
                                    hipMemcpyHostToDevice, ctx->stream));
         if (L.ne2) {
@@ -930,8 +930,8 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
                                    hipMemcpyHostToDevice, ctx->stream));
         (void)hipFree(d_comm);
         (void)hipFree(d_csize);
-        MGX_HIP_TRY(hipMalloc(&d_comm, n_new * 4));
-        MGX_HIP_TRY(hipMalloc(&d_csize, n_new * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&d_comm, n_new * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&d_csize, n_new * 4));
         MGX_HIP_TRY(hipMemcpyAsync(d_comm, comm_new.data(), n_new * 4,
                                    hipMemcpyHostToDevice, ctx->stream));
         std::vector<uint32_t> cs_new((size_t)n_new, 0);

@@ -1267,7 +1267,7 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
 
   out->nv = n_clusters;
   ctx->ensure_margin((size_t)(n_clusters + 1) * 4);
-  MGX_HIP_TRY(hipMalloc(&out->row_ptr, (n_clusters + 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&out->row_ptr, (n_clusters + 1) * 4));
   if (louvain_trace()) {
     // Pre-scan probe: counts must still hold the pair histogram, and the
     // freshly hipMalloc'd row_ptr must not alias the pool-allocated counts
@@ -1330,8 +1330,8 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
     return MGX_ERR_HIP;
   }
   ctx->ensure_margin((size_t)(total > 0 ? total : 1) * 12);
-  MGX_HIP_TRY(hipMalloc(&out->col, (total > 0 ? total : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&out->w, (total > 0 ? total : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&out->col, (total > 0 ? total : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&out->w, (total > 0 ? total : 1) * 8));
   if (louvain_trace()) {
     // Allocator-overlap probe: the async-pool DevBufs vs the plain hipMalloc
     // level buffers must be disjoint VA ranges.

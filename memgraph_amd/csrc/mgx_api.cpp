@@ -65,7 +65,7 @@ size_t round_bucket(size_t bytes) {
 
 mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
   if (no_alloc_cache()) {
-    MGX_HIP_TRY(hipMalloc(p, bytes ? bytes : 1));
+    MGX_HIP_TRY(mgx_hip_malloc(p, bytes ? bytes : 1));
     return MGX_OK;
   }
   const size_t want = round_bucket(bytes);
@@ -82,15 +82,15 @@ mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
     }
   }
   ensure_margin(want);
-  hipError_t e = hipMalloc(p, want);
+  hipError_t e = mgx_hip_malloc(p, want);
   if (e == hipErrorOutOfMemory) {
     (void)hipGetLastError();
     cache_trim();  // return every cached free block, then retry once
-    e = hipMalloc(p, want);
+    e = mgx_hip_malloc(p, want);
   }
   if (e != hipSuccess) {
     (void)hipGetLastError();
-    mgx_set_error("hipMalloc(%zu) failed: %s", want, hipGetErrorString(e));
+    mgx_set_error("mgx_hip_malloc(%zu) failed: %s", want, hipGetErrorString(e));
     return e == hipErrorOutOfMemory ? MGX_ERR_OUT_OF_MEMORY : MGX_ERR_HIP;
   }
   std::lock_guard<std::mutex> lk(cache_mu);
@@ -320,8 +320,8 @@ extern "C" mgx_status mgx_graph_rmat_sharded(mgx_context *ctx, int scale, int64_
   }
   MGX_HIP_TRY(hipSetDevice(ctx->device));
   int32_t *d_src = nullptr, *d_dst = nullptr;
-  MGX_HIP_TRY(hipMalloc(&d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(hipMalloc(&d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
   mgx_status s = mgx_gen_rmat_device(ctx, scale, n_edges, seed, a, b, c, d_src, d_dst);
   mgx_graph *g = nullptr;
   if (s == MGX_OK) {

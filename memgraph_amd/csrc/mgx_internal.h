@@ -37,6 +37,24 @@ void mgx_set_error(const char *fmt, ...);
 
 struct mgx_comm_state;  // comm.cpp (RCCL)
 
+// The box's ROCm 7.0.x runtime mis-reserves plain hipMalloc requests that
+// are >4 GiB and NOT a multiple of 2^32 bytes: the returned range overlaps
+// live allocations and writing past the low 32 bits of the size faults
+// ("write to a read-only page"). Every multi-GB allocation that ever
+// worked here was an exact 2^32 multiple (power-of-two element counts);
+// the first non-aligned huge malloc (Louvain coarsen's total-sized w at
+// RMAT-25) hit the bug. Round such sizes up to the next 2^32 multiple.
+inline size_t mgx_safe_size(size_t bytes) {
+  if (bytes >= (4ull << 30) && (bytes & 0xFFFFFFFFull) != 0)
+    bytes = (bytes + 0xFFFFFFFFull) & ~0xFFFFFFFFull;
+  return bytes;
+}
+
+template <typename T>
+inline hipError_t mgx_hip_malloc(T **p, size_t bytes) {
+  return hipMalloc((void **)p, mgx_safe_size(bytes));
+}
+
 struct mgx_context {
   int device = -1;
   hipStream_t stream = nullptr;

@@ -257,9 +257,9 @@ mgx_status ensure_pool(mgx_context *ctx, int64_t need_cap) {
   while (cap < need_cap) cap *= 2;
   uint32_t *nw = nullptr, *np = nullptr;
   int32_t *nn = nullptr;
-  MGX_HIP_TRY(hipMalloc(&nw, cap * 4));
-  MGX_HIP_TRY(hipMalloc(&np, cap * 4));
-  MGX_HIP_TRY(hipMalloc(&nn, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nw, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&np, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nn, cap * 4));
   if (g_st.pool_used > 0) {
     MGX_HIP_TRY(hipMemcpyAsync(nw, g_st.e_walk, g_st.pool_used * 4,
                                hipMemcpyDeviceToDevice, ctx->stream));
@@ -286,9 +286,9 @@ mgx_status ensure_walks(mgx_context *ctx, int64_t need) {
   int32_t *ns = nullptr;
   uint32_t *ng = nullptr;
   uint8_t *nd = nullptr;
-  MGX_HIP_TRY(hipMalloc(&ns, cap * 4));
-  MGX_HIP_TRY(hipMalloc(&ng, cap * 4));
-  MGX_HIP_TRY(hipMalloc(&nd, cap));
+  MGX_HIP_TRY(mgx_hip_malloc(&ns, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&ng, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nd, cap));
   MGX_HIP_TRY(hipMemsetAsync(ng, 0, cap * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(nd, 0, cap, ctx->stream));
   if (g_st.n_walks > 0) {
@@ -316,7 +316,7 @@ mgx_status sync_slot_alive(mgx_context *ctx) {
     if (g_st.d_slot_alive) (void)hipFree(g_st.d_slot_alive);
     int64_t cap = 64;
     while (cap < n) cap *= 2;
-    MGX_HIP_TRY(hipMalloc(&g_st.d_slot_alive, cap));
+    MGX_HIP_TRY(mgx_hip_malloc(&g_st.d_slot_alive, cap));
     g_st.slots_cap_dev = cap;
   }
   if (n > 0) {
@@ -385,8 +385,8 @@ mgx_status build_maps(mgx_context *ctx, const int64_t *dense_to_mg, int64_t V, b
   std::vector<int32_t> s2d(n_slots > 0 ? n_slots : 1, -1);
   for (int64_t v = 0; v < V; ++v)
     if (d2s[v] >= 0) s2d[d2s[v]] = (int32_t)v;
-  MGX_HIP_TRY(hipMalloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
-  MGX_HIP_TRY(hipMalloc(&m->slot2dense, (n_slots > 0 ? n_slots : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->slot2dense, (n_slots > 0 ? n_slots : 1) * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m->dense2slot, d2s.data(), (V > 0 ? V : 1) * 4,
                              hipMemcpyHostToDevice, ctx->stream));
   MGX_HIP_TRY(hipMemcpyAsync(m->slot2dense, s2d.data(), (n_slots > 0 ? n_slots : 1) * 4,
@@ -404,7 +404,7 @@ mgx_status run_gen(mgx_context *ctx, const Maps &m, mgx_graph *g,
                    int64_t expected_append) {
   if (n_items == 0) return MGX_OK;
   if (!g_st.d_cursor) {
-    MGX_HIP_TRY(hipMalloc(&g_st.d_cursor, 16));
+    MGX_HIP_TRY(mgx_hip_malloc(&g_st.d_cursor, 16));
   }
   uint64_t eps_bits;
   if (eps >= 1.0) {
