@@ -60,27 +60,5 @@ int main() {
     ++idx;
   }
   printf("DONE\n");
-
-  // Hole repro: free a ~525 MB allocation to punch a VA hole below a live
-  // block, then request >4 GiB. On the broken runtime the fit check
-  // compares only (size mod 2^32), so the big block lands in the hole and
-  // overlaps everything above it.
-  for (auto &l : live) TRY(hipFree(l.first));
-  live.clear();
-  void *hole = nullptr, *keeper = nullptr, *big = nullptr;
-  TRY(hipMalloc(&hole, 525336576));      // to be freed -> hole
-  TRY(hipMalloc(&keeper, 2147483648ull));  // stays live above/below the hole
-  printf("hole=%p..%p keeper=%p..%p\n", hole, (char*)hole + 525336576, keeper,
-         (char*)keeper + 2147483648ull);
-  TRY(hipFree(hole));
-  hipError_t e2 = hipMalloc(&big, 4817158472ull);
-  printf("big(4.82GB after hole) -> %p..%p (%s)\n", big,
-         (char*)big + 4817158472ull, hipGetErrorString(e2));
-  if (e2 == hipSuccess) {
-    const char *a0 = (const char*)big, *a1 = a0 + 4817158472ull;
-    const char *b0 = (const char*)keeper, *b1 = b0 + 2147483648ull;
-    printf("%s\n", (a0 < b1 && b0 < a1) ? "HOLE-BUG: big overlaps keeper"
-                                         : "hole fit OK");
-  }
   return 0;
 }

@@ -384,13 +384,13 @@ mgx_status mgx_build_bins_range(mgx_context *ctx, const uint32_t *lo, const uint
     bins->rows = nullptr;
     return MGX_OK;
   }
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &bins->rows, rows * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&bins->rows, rows * sizeof(int32_t)));
   uint32_t *keys = nullptr, *keys_out = nullptr, *vals = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &keys, rows * sizeof(uint32_t)));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &keys_out, rows * sizeof(uint32_t)));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &vals, rows * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&keys, rows * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&keys_out, rows * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&vals, rows * sizeof(uint32_t)));
   uint32_t *counts5 = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &counts5, 5 * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&counts5, 5 * sizeof(uint32_t)));
   MGX_HIP_TRY(hipMemsetAsync(counts5, 0, 20, ctx->stream));
 
   hipLaunchKernelGGL(k_bin_keys, dim3(grid_for(rows)), dim3(kBlock), 0, ctx->stream, rows,
@@ -424,10 +424,10 @@ mgx_status mgx_build_bins_range(mgx_context *ctx, const uint32_t *lo, const uint
     bins->grid[b] = bins->count[b] ? (need < cap[b] ? need : cap[b]) : 0;
   }
 
-  MGX_TRY(mgx_ctx_free_raw(ctx, keys));
-  MGX_TRY(mgx_ctx_free_raw(ctx, keys_out));
-  MGX_TRY(mgx_ctx_free_raw(ctx, vals));
-  MGX_TRY(mgx_ctx_free_raw(ctx, counts5));
+  MGX_HIP_TRY(hipFree(keys));
+  MGX_HIP_TRY(hipFree(keys_out));
+  MGX_HIP_TRY(hipFree(vals));
+  MGX_HIP_TRY(hipFree(counts5));
   return MGX_OK;
 }
 
@@ -480,7 +480,7 @@ mgx_status mgx_build_stripes(mgx_context *ctx, mgx_graph *g) {
   if (n_stripes == 1) return MGX_OK;
   g->stripe_width = (V + n_stripes - 1) / n_stripes;
 
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->stripe_ptr, (size_t)(n_stripes + 1) * rows * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->stripe_ptr, (size_t)(n_stripes + 1) * rows * sizeof(uint32_t)));
   // boundaries: sp[0] = row starts, sp[S] = row ends, sp[s] = searchsorted.
   struct CopyK {
     static __global__ void shift(int64_t rows, const uint32_t *row_ptr, uint32_t *lo,
@@ -527,14 +527,14 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
   MGX_HIP_TRY(hipEventRecord(ev0, ctx->stream));
 
   uint32_t *counts = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &counts, (V > 0 ? V : 1) * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&counts, (V > 0 ? V : 1) * sizeof(uint32_t)));
 
   // out-degree (+ inv) — always needed by PageRank/Katz result paths.
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->out_degree, (V > 0 ? V : 1) * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->out_degree, (V > 0 ? V : 1) * sizeof(uint32_t)));
   MGX_HIP_TRY(hipMemsetAsync(g->out_degree, 0, V * sizeof(uint32_t), ctx->stream));
   hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
                      g->out_degree);
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->inv_outdeg, (V > 0 ? V : 1) * sizeof(float)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->inv_outdeg, (V > 0 ? V : 1) * sizeof(float)));
   hipLaunchKernelGGL(k_inv_outdeg, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                      g->out_degree, g->inv_outdeg);
 
@@ -543,11 +543,11 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
     hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_dst,
                        counts);
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
     MGX_TRY(scan_counts(ctx, counts, V, g->in_row_ptr));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
     if ((flags & MGX_BUILD_WEIGHTED) && d_w) {
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_w, (E > 0 ? E : 1) * sizeof(float)));
+      MGX_HIP_TRY(mgx_hip_malloc(&g->in_w, (E > 0 ? E : 1) * sizeof(float)));
       MGX_TRY(build_sorted_cols_w(ctx, d_src, d_dst, d_w, E, V, g->in_col, g->in_w));
     } else {
       MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, nullptr, E, V, false, 0, 0, g->in_col,
@@ -559,13 +559,13 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     // (stable, ties by original id) so the most-gathered contrib entries
     // pack into the lowest addresses (L2/L3-resident under power-law skew).
     int32_t *d_perm = nullptr;
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->order, (V > 0 ? V : 1) * sizeof(int32_t)));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_perm, (V > 0 ? V : 1) * sizeof(int32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->order, (V > 0 ? V : 1) * sizeof(int32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_perm, (V > 0 ? V : 1) * sizeof(int32_t)));
     {
       uint32_t *deg_sorted = nullptr;
       int32_t *iota = nullptr;
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &deg_sorted, (V > 0 ? V : 1) * sizeof(uint32_t)));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &iota, (V > 0 ? V : 1) * sizeof(int32_t)));
+      MGX_HIP_TRY(mgx_hip_malloc(&deg_sorted, (V > 0 ? V : 1) * sizeof(uint32_t)));
+      MGX_HIP_TRY(mgx_hip_malloc(&iota, (V > 0 ? V : 1) * sizeof(int32_t)));
       hipLaunchKernelGGL(k_iota_i32g, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                          iota);
       size_t tmp_bytes = 0;
@@ -585,17 +585,17 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
                                  hipMemcpyDeviceToDevice, ctx->stream));
       hipLaunchKernelGGL(k_inv_outdeg, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                          g->out_degree, g->inv_outdeg);
-      MGX_TRY(mgx_ctx_free_raw(ctx, deg_sorted));
-      MGX_TRY(mgx_ctx_free_raw(ctx, iota));
+      MGX_HIP_TRY(hipFree(deg_sorted));
+      MGX_HIP_TRY(hipFree(iota));
     }
     MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
     hipLaunchKernelGGL(k_hist_perm, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
                        d_dst, d_perm, counts);
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->in_row_ptr, (V + 1) * sizeof(uint32_t)));
     MGX_TRY(scan_counts(ctx, counts, V, g->in_row_ptr));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->in_col, (E > 0 ? E : 1) * sizeof(int32_t)));
     MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, d_perm, E, V, false, 0, 0, g->in_col, E));
-    MGX_TRY(mgx_ctx_free_raw(ctx, d_perm));
+    MGX_HIP_TRY(hipFree(d_perm));
     MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, V, &g->bins_in));
     MGX_TRY(mgx_build_stripes(ctx, g));
   }
@@ -606,9 +606,9 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
     hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
                        counts);
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->out_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->out_row_ptr, (V + 1) * sizeof(uint32_t)));
     MGX_TRY(scan_counts(ctx, counts, V, g->out_row_ptr));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->out_col, (E > 0 ? E : 1) * sizeof(int32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->out_col, (E > 0 ? E : 1) * sizeof(int32_t)));
     // reuse the sorted-cols builder with (src,dst) swapped: rows = sources
     MGX_TRY(build_sorted_cols(ctx, d_dst, d_src, nullptr, E, V, false, 0, 0, g->out_col,
                               E));
@@ -619,17 +619,17 @@ mgx_status mgx_build_from_device_coo(mgx_context *ctx, const int32_t *d_src,
     MGX_HIP_TRY(hipMemsetAsync(counts, 0, V * sizeof(uint32_t), ctx->stream));
     hipLaunchKernelGGL(k_hist2, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
                        d_dst, counts);
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->sym_row_ptr, (V + 1) * sizeof(uint32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->sym_row_ptr, (V + 1) * sizeof(uint32_t)));
     MGX_TRY(scan_counts(ctx, counts, V, g->sym_row_ptr));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->sym_col, (E > 0 ? 2 * E : 1) * sizeof(int32_t)));
+    MGX_HIP_TRY(mgx_hip_malloc(&g->sym_col, (E > 0 ? 2 * E : 1) * sizeof(int32_t)));
     if (flags & MGX_BUILD_WEIGHTED) {
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->sym_w, (E > 0 ? 2 * E : 1) * sizeof(float)));
+      MGX_HIP_TRY(mgx_hip_malloc(&g->sym_w, (E > 0 ? 2 * E : 1) * sizeof(float)));
     }
     MGX_TRY(build_sorted_sym(ctx, d_src, d_dst, d_w, E, V, g->sym_col, g->sym_w));
     MGX_TRY(mgx_build_bins(ctx, g->sym_row_ptr, V, &g->bins_sym));
   }
 
-  MGX_TRY(mgx_ctx_free_raw(ctx, counts));
+  MGX_HIP_TRY(hipFree(counts));
   MGX_HIP_TRY(hipEventRecord(ev1, ctx->stream));
   MGX_HIP_TRY(hipEventSynchronize(ev1));
   float ms = 0.f;
@@ -659,32 +659,32 @@ mgx_status mgx_build_sharded_in_csr(mgx_context *ctx, const int32_t *d_src,
   MGX_HIP_TRY(hipEventRecord(ev0, ctx->stream));
 
   // Global out-degree (contrib denominators need every source).
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->out_degree, V * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->out_degree, V * sizeof(uint32_t)));
   MGX_HIP_TRY(hipMemsetAsync(g->out_degree, 0, V * sizeof(uint32_t), ctx->stream));
   hipLaunchKernelGGL(k_hist, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E, d_src,
                      g->out_degree);
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->inv_outdeg, V * sizeof(float)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->inv_outdeg, V * sizeof(float)));
   hipLaunchKernelGGL(k_inv_outdeg, dim3(grid_for(V)), dim3(kBlock), 0, ctx->stream, V,
                      g->out_degree, g->inv_outdeg);
 
   uint32_t *counts = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &counts, (rows > 0 ? rows : 1) * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&counts, (rows > 0 ? rows : 1) * sizeof(uint32_t)));
   MGX_HIP_TRY(hipMemsetAsync(counts, 0, rows * sizeof(uint32_t), ctx->stream));
   hipLaunchKernelGGL(k_hist_ranged, dim3(grid_for(E)), dim3(kBlock), 0, ctx->stream, E,
                      d_dst, (int32_t)row_begin, (int32_t)row_end, counts);
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_row_ptr, (rows + 1) * sizeof(uint32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->in_row_ptr, (rows + 1) * sizeof(uint32_t)));
   MGX_TRY(scan_counts(ctx, counts, rows, g->in_row_ptr));
   uint32_t local_edges = 0;
   MGX_HIP_TRY(hipMemcpyAsync(&local_edges, g->in_row_ptr + rows, 4, hipMemcpyDeviceToHost,
                              ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   g->in_edges = local_edges;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g->in_col, (local_edges > 0 ? local_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&g->in_col, (local_edges > 0 ? local_edges : 1) * sizeof(int32_t)));
   MGX_TRY(build_sorted_cols(ctx, d_src, d_dst, nullptr, E, rows, true, (int32_t)row_begin,
                             (int32_t)row_end, g->in_col, local_edges));
   MGX_TRY(mgx_build_bins(ctx, g->in_row_ptr, rows, &g->bins_in));
   MGX_TRY(mgx_build_stripes(ctx, g));
-  MGX_TRY(mgx_ctx_free_raw(ctx, counts));
+  MGX_HIP_TRY(hipFree(counts));
 
   MGX_HIP_TRY(hipEventRecord(ev1, ctx->stream));
   MGX_HIP_TRY(hipEventSynchronize(ev1));
@@ -701,10 +701,10 @@ mgx_status mgx_build_sharded_in_csr(mgx_context *ctx, const int32_t *d_src,
 mgx_status mgx_upload_coo(mgx_context *ctx, const int64_t *src, const int64_t *dst,
                           const double *weights, int64_t n_edges, int32_t **d_src,
                           int32_t **d_dst, float **d_w) {
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
   *d_w = nullptr;
-  if (weights) MGX_HIP_TRY(mgx_ctx_malloc(ctx, d_w, (n_edges > 0 ? n_edges : 1) * sizeof(float)));
+  if (weights) MGX_HIP_TRY(mgx_hip_malloc(d_w, (n_edges > 0 ? n_edges : 1) * sizeof(float)));
 
   // Chunked staging: int64 -> int32 converted on device.
   const int64_t chunk = 16 << 20;
@@ -712,8 +712,8 @@ mgx_status mgx_upload_coo(mgx_context *ctx, const int64_t *src, const int64_t *d
   double *stagef = nullptr;
   const int64_t this_chunk = n_edges < chunk ? n_edges : chunk;
   if (this_chunk > 0) {
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &stage64, this_chunk * sizeof(int64_t)));
-    if (weights) MGX_HIP_TRY(mgx_ctx_malloc(ctx, &stagef, this_chunk * sizeof(double)));
+    MGX_HIP_TRY(mgx_hip_malloc(&stage64, this_chunk * sizeof(int64_t)));
+    if (weights) MGX_HIP_TRY(mgx_hip_malloc(&stagef, this_chunk * sizeof(double)));
   }
   for (int64_t off = 0; off < n_edges; off += chunk) {
     const int64_t n = (n_edges - off) < chunk ? (n_edges - off) : chunk;
@@ -734,8 +734,8 @@ mgx_status mgx_upload_coo(mgx_context *ctx, const int64_t *src, const int64_t *d
     // The next chunk reuses the staging buffer: wait for queued copies.
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   }
-  if (stage64) MGX_TRY(mgx_ctx_free_raw(ctx, stage64));
-  if (stagef) MGX_TRY(mgx_ctx_free_raw(ctx, stagef));
+  if (stage64) MGX_HIP_TRY(hipFree(stage64));
+  if (stagef) MGX_HIP_TRY(hipFree(stagef));
   return MGX_OK;
 }
 
@@ -756,9 +756,9 @@ extern "C" mgx_status mgx_gen_edges_to_host(mgx_context *ctx, int rmat, int scal
   MGX_HIP_TRY(hipSetDevice(ctx->device));
   int32_t *d_src = nullptr, *d_dst = nullptr;
   int64_t *d_wide = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_wide, (n_edges > 0 ? n_edges : 1) * sizeof(int64_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_wide, (n_edges > 0 ? n_edges : 1) * sizeof(int64_t)));
   mgx_status s;
   if (rmat) {
     s = mgx_gen_rmat_device(ctx, scale, n_edges, seed, a, b, c, d_src, d_dst);
@@ -777,8 +777,8 @@ extern "C" mgx_status mgx_gen_edges_to_host(mgx_context *ctx, int rmat, int scal
                          ctx->stream);
     (void)hipStreamSynchronize(ctx->stream);
   }
-  mgx_ctx_free(ctx, d_src);
-  mgx_ctx_free(ctx, d_dst);
-  mgx_ctx_free(ctx, d_wide);
+  (void)hipFree(d_src);
+  (void)hipFree(d_dst);
+  (void)hipFree(d_wide);
   return s;
 }

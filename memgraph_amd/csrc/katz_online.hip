@@ -292,12 +292,12 @@ __global__ void k_gather_out(int64_t V, const int32_t *dense2slot, const double 
 
 // ---- host helpers --------------------------------------------------------
 
-void kon_free_state(mgx_context *ctx) {
-  for (auto p : g_k.omega) mgx_ctx_free(ctx, p);
-  for (auto p : g_k.cent) mgx_ctx_free(ctx, p);
-  if (g_k.lr) mgx_ctx_free(ctx, g_k.lr);
-  if (g_k.ur) mgx_ctx_free(ctx, g_k.ur);
-  if (g_k.d_active) mgx_ctx_free(ctx, g_k.d_active);
+void kon_free_state() {
+  for (auto p : g_k.omega) (void)hipFree(p);
+  for (auto p : g_k.cent) (void)hipFree(p);
+  if (g_k.lr) (void)hipFree(g_k.lr);
+  if (g_k.ur) (void)hipFree(g_k.ur);
+  if (g_k.d_active) (void)hipFree(g_k.d_active);
   g_k = KOnState{};
 }
 
@@ -318,7 +318,7 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
   while (cap < need) cap *= 2;
   auto grow = [&](double **p, double init) -> mgx_status {
     double *np = nullptr;
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &np, cap * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&np, cap * 8));
     hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(cap)), dim3(kBlock), 0,
                        ctx->stream, cap, init, np);
     if (*p && g_k.slots_cap > 0) {
@@ -327,7 +327,7 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
     }
     if (*p) {
       MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-      mgx_ctx_free(ctx, *p);
+      (void)hipFree(*p);
     }
     *p = np;
     return MGX_OK;
@@ -339,14 +339,14 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
   MGX_TRY(grow(&g_k.ur, 0.0));
   {
     uint8_t *na = nullptr;
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &na, cap));
+    MGX_HIP_TRY(mgx_hip_malloc(&na, cap));
     MGX_HIP_TRY(hipMemsetAsync(na, 0, cap, ctx->stream));
     if (g_k.d_active && g_k.slots_cap > 0) {
       MGX_HIP_TRY(hipMemcpyAsync(na, g_k.d_active, g_k.slots_cap,
                                  hipMemcpyDeviceToDevice, ctx->stream));
     }
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-    if (g_k.d_active) mgx_ctx_free(ctx, g_k.d_active);
+    if (g_k.d_active) (void)hipFree(g_k.d_active);
     g_k.d_active = na;
   }
   g_k.slots_cap = cap;
@@ -355,8 +355,8 @@ mgx_status kon_grow_slots(mgx_context *ctx, int64_t need) {
 
 mgx_status kon_add_iteration(mgx_context *ctx, double omega_init) {
   double *w = nullptr, *c = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &w, g_k.slots_cap * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &c, g_k.slots_cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&w, g_k.slots_cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&c, g_k.slots_cap * 8));
   hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(g_k.slots_cap)), dim3(kBlock), 0,
                      ctx->stream, g_k.slots_cap, omega_init, w);
   hipLaunchKernelGGL(k_fill_f64, dim3((uint32_t)grid_for(g_k.slots_cap)), dim3(kBlock), 0,
@@ -367,7 +367,6 @@ mgx_status kon_add_iteration(mgx_context *ctx, double omega_init) {
 }
 
 struct KMaps {
-  mgx_context *ctx = nullptr;  // for the caching allocator in ~KMaps
   int32_t *dense2slot = nullptr;  // device
   std::vector<int32_t> d2s_h;
   int64_t V = 0;
@@ -375,17 +374,15 @@ struct KMaps {
   int32_t *small_rows = nullptr, *big_rows = nullptr;
   int64_t n_small = 0, n_big = 0;
   ~KMaps() {
-    if (!ctx) return;
-    if (dense2slot) mgx_ctx_free(ctx, dense2slot);
-    if (small_rows) mgx_ctx_free(ctx, small_rows);
-    if (big_rows) mgx_ctx_free(ctx, big_rows);
+    if (dense2slot) (void)hipFree(dense2slot);
+    if (small_rows) (void)hipFree(small_rows);
+    if (big_rows) (void)hipFree(big_rows);
   }
 };
 
 mgx_status kon_build_maps(mgx_context *ctx, mgx_graph *g, const int64_t *dense_to_mg,
                           bool create, KMaps *m, bool *all_known) {
   const int64_t V = g ? g->n_vertices : 0;
-  m->ctx = ctx;
   m->V = V;
   m->d2s_h.assign(V > 0 ? V : 1, -1);
   bool known = true;
@@ -403,7 +400,7 @@ mgx_status kon_build_maps(mgx_context *ctx, mgx_graph *g, const int64_t *dense_t
     }
   }
   if (all_known) *all_known = known;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m->dense2slot, (V > 0 ? V : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m->dense2slot, m->d2s_h.data(), (V > 0 ? V : 1) * 4,
                              hipMemcpyHostToDevice, ctx->stream));
   // in-degree bins from the in-CSR (host-side row_ptr copy)
@@ -420,8 +417,8 @@ mgx_status kon_build_maps(mgx_context *ctx, mgx_graph *g, const int64_t *dense_t
     }
     m->n_small = (int64_t)sm.size();
     m->n_big = (int64_t)bg.size();
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m->small_rows, (m->n_small > 0 ? m->n_small : 1) * 4));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m->big_rows, (m->n_big > 0 ? m->n_big : 1) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&m->small_rows, (m->n_small > 0 ? m->n_small : 1) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&m->big_rows, (m->n_big > 0 ? m->n_big : 1) * 4));
     // empty vectors have a null data(): copy only when non-empty
     if (m->n_small > 0)
       MGX_HIP_TRY(hipMemcpyAsync(m->small_rows, sm.data(), m->n_small * 4,
@@ -446,11 +443,11 @@ mgx_status kon_loop(mgx_context *ctx, mgx_graph *g, const KMaps &m, double gamma
   int32_t *d_act = nullptr, *d_act_sorted = nullptr;
   uint64_t *d_keys = nullptr, *d_keys_sorted = nullptr;
   uint32_t *d_flag = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_act, (n_active > 0 ? n_active : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_act_sorted, (n_active > 0 ? n_active : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_keys, (n_active > 0 ? n_active : 1) * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_keys_sorted, (n_active > 0 ? n_active : 1) * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_flag, 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_act, (n_active > 0 ? n_active : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_act_sorted, (n_active > 0 ? n_active : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_keys, (n_active > 0 ? n_active : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_keys_sorted, (n_active > 0 ? n_active : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_flag, 4));
   if (n_active > 0)
     MGX_HIP_TRY(hipMemcpyAsync(d_act, act.data(), n_active * 4, hipMemcpyHostToDevice,
                                ctx->stream));
@@ -512,11 +509,11 @@ mgx_status kon_loop(mgx_context *ctx, mgx_graph *g, const KMaps &m, double gamma
       break;
     }
   }
-  mgx_ctx_free(ctx, d_act);
-  mgx_ctx_free(ctx, d_act_sorted);
-  mgx_ctx_free(ctx, d_keys);
-  mgx_ctx_free(ctx, d_keys_sorted);
-  mgx_ctx_free(ctx, d_flag);
+  (void)hipFree(d_act);
+  (void)hipFree(d_act_sorted);
+  (void)hipFree(d_keys);
+  (void)hipFree(d_keys_sorted);
+  (void)hipFree(d_flag);
   (void)V;
   return st;
 }
@@ -557,7 +554,7 @@ extern "C" int mgx_konline_initialized(void) {
 
 extern "C" mgx_status mgx_konline_reset(mgx_context *ctx) {
   (void)ctx;
-  kon_free_state(ctx);
+  kon_free_state();
   return MGX_OK;
 }
 
@@ -572,7 +569,7 @@ extern "C" mgx_status mgx_konline_set(mgx_context *ctx, mgx_graph *g,
     return MGX_ERR_INVALID_ARGUMENT;
   }
   MGX_HIP_TRY(hipSetDevice(ctx->device));
-  kon_free_state(ctx);
+  kon_free_state();
   g_k.alpha = alpha;
   g_k.eps = epsilon;
   g_k.initialized = true;
@@ -614,7 +611,7 @@ extern "C" mgx_status mgx_konline_get(mgx_context *ctx, const int64_t *dense_to_
   if (!ok || !out || V == 0) return MGX_OK;
   KMaps m;
   m.V = V;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m.dense2slot, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m.dense2slot, V * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m.dense2slot, d2s.data(), V * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   return kon_output(ctx, m, out);
@@ -700,8 +697,8 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
   }
   uint64_t *d_pairs = nullptr;
   uint32_t *d_cnt = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_pairs, (pair_keys.empty() ? 1 : pair_keys.size()) * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_cnt, (pair_cnt.empty() ? 1 : pair_cnt.size()) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_pairs, (pair_keys.empty() ? 1 : pair_keys.size()) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_cnt, (pair_cnt.empty() ? 1 : pair_cnt.size()) * 4));
   if (!pair_keys.empty()) {
     MGX_HIP_TRY(hipMemcpyAsync(d_pairs, pair_keys.data(), pair_keys.size() * 8,
                                hipMemcpyHostToDevice, ctx->stream));
@@ -726,7 +723,7 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
     del_vs.push_back(v->second);
   }
   auto upload_i32 = [&](const std::vector<int32_t> &v, int32_t **d) -> mgx_status {
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, d, (v.empty() ? 1 : v.size()) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(d, (v.empty() ? 1 : v.size()) * 4));
     if (!v.empty())
       MGX_HIP_TRY(hipMemcpyAsync(*d, v.data(), v.size() * 4, hipMemcpyHostToDevice,
                                  ctx->stream));
@@ -744,10 +741,10 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
   std::vector<double *> new_omega(g_k.iteration + 1, nullptr);
   auto cleanup_new = [&]() {
     for (auto p : new_omega)
-      if (p) mgx_ctx_free(ctx, p);
+      if (p) (void)hipFree(p);
   };
   for (int64_t i = 0; i <= g_k.iteration; ++i) {
-    if (mgx_ctx_malloc(ctx, &new_omega[i], g_k.slots_cap * 8) != hipSuccess) {
+    if (mgx_hip_malloc(&new_omega[i], g_k.slots_cap * 8) != hipSuccess) {
       cleanup_new();
       mgx_set_error("konline_update: out of memory for level omegas");
       return MGX_ERR_OUT_OF_MEMORY;
@@ -760,10 +757,10 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
   uint32_t *d_updated = nullptr;
   int32_t *d_front = nullptr, *d_next = nullptr;
   unsigned long long *d_nn = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_updated, V * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_front, V * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_next, V * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_nn, 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_updated, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_front, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_next, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_nn, 8));
   std::vector<uint32_t> upd_h(V, 0);
   std::vector<int32_t> front_h;
   for (auto mg : updated_mg) {
@@ -845,7 +842,7 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
                        d_upd_slots, lvl, pow(g_k.alpha, (double)lvl), new_omega[lvl],
                        g_k.omega[lvl], g_k.cent[lvl - 1], g_k.cent[lvl]);
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-    mgx_ctx_free(ctx, d_upd_slots);
+    (void)hipFree(d_upd_slots);
   }
 
   // merge new omegas into state (:427-431)
@@ -868,7 +865,7 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
                        g_k.cent[g_k.iteration], g_k.omega[g_k.iteration],
                        pow(g_k.alpha, (double)(g_k.iteration + 1)), g_k.lr, g_k.ur);
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-    mgx_ctx_free(ctx, d_upd_slots);
+    (void)hipFree(d_upd_slots);
   }
 
   // re-activation (:442-455): min lr over active, then ur >= min_lr - eps
@@ -895,16 +892,16 @@ extern "C" mgx_status mgx_konline_update(mgx_context *ctx, mgx_graph *g,
     g_k.active_h[it->second] = 0;
   }
 
-  mgx_ctx_free(ctx, d_pairs);
-  mgx_ctx_free(ctx, d_cnt);
-  mgx_ctx_free(ctx, d_new_ws);
-  mgx_ctx_free(ctx, d_new_vs);
-  mgx_ctx_free(ctx, d_del_ws);
-  mgx_ctx_free(ctx, d_del_vs);
-  mgx_ctx_free(ctx, d_updated);
-  mgx_ctx_free(ctx, d_front);
-  mgx_ctx_free(ctx, d_next);
-  mgx_ctx_free(ctx, d_nn);
+  (void)hipFree(d_pairs);
+  (void)hipFree(d_cnt);
+  (void)hipFree(d_new_ws);
+  (void)hipFree(d_new_vs);
+  (void)hipFree(d_del_ws);
+  (void)hipFree(d_del_vs);
+  (void)hipFree(d_updated);
+  (void)hipFree(d_front);
+  (void)hipFree(d_next);
+  (void)hipFree(d_nn);
 
   MGX_TRY(kon_loop(ctx, g, m, gamma));
   return kon_output(ctx, m, out);

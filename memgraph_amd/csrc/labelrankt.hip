@@ -387,13 +387,13 @@ __global__ void k_argmax(int64_t n_slots, const int32_t *lab, const double *p,
 
 namespace {
 
-void lrt_free(mgx_context *ctx) {
-  if (g_l.lab) mgx_ctx_free(ctx, g_l.lab);
-  if (g_l.p) mgx_ctx_free(ctx, g_l.p);
-  if (g_l.off) mgx_ctx_free(ctx, g_l.off);
-  if (g_l.len) mgx_ctx_free(ctx, g_l.len);
-  if (g_l.sum_w) mgx_ctx_free(ctx, g_l.sum_w);
-  if (g_l.times_upd) mgx_ctx_free(ctx, g_l.times_upd);
+void lrt_free() {
+  if (g_l.lab) (void)hipFree(g_l.lab);
+  if (g_l.p) (void)hipFree(g_l.p);
+  if (g_l.off) (void)hipFree(g_l.off);
+  if (g_l.len) (void)hipFree(g_l.len);
+  if (g_l.sum_w) (void)hipFree(g_l.sum_w);
+  if (g_l.times_upd) (void)hipFree(g_l.times_upd);
   g_l = LrtState{};
 }
 
@@ -414,10 +414,10 @@ mgx_status lrt_grow_slots(mgx_context *ctx, int64_t need) {
   uint64_t *noff = nullptr;
   uint32_t *nlen = nullptr, *ntu = nullptr;
   double *nsw = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &noff, cap * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nlen, cap * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &ntu, cap * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nsw, cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&noff, cap * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&nlen, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&ntu, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nsw, cap * 8));
   MGX_HIP_TRY(hipMemsetAsync(nlen, 0, cap * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(ntu, 0, cap * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(noff, 0, cap * 8, ctx->stream));
@@ -432,10 +432,10 @@ mgx_status lrt_grow_slots(mgx_context *ctx, int64_t need) {
                                hipMemcpyDeviceToDevice, ctx->stream));
   }
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  if (g_l.off) mgx_ctx_free(ctx, g_l.off);
-  if (g_l.len) mgx_ctx_free(ctx, g_l.len);
-  if (g_l.times_upd) mgx_ctx_free(ctx, g_l.times_upd);
-  if (g_l.sum_w) mgx_ctx_free(ctx, g_l.sum_w);
+  if (g_l.off) (void)hipFree(g_l.off);
+  if (g_l.len) (void)hipFree(g_l.len);
+  if (g_l.times_upd) (void)hipFree(g_l.times_upd);
+  if (g_l.sum_w) (void)hipFree(g_l.sum_w);
   g_l.off = noff;
   g_l.len = nlen;
   g_l.times_upd = ntu;
@@ -465,23 +465,21 @@ mgx_status scan_counts_u64(mgx_context *ctx, const uint32_t *d_counts, int64_t n
 }
 
 struct LrtMaps {
-  mgx_context *ctx = nullptr;  // for the caching allocator in ~LrtMaps
   int32_t *dense2slot = nullptr;
   std::vector<int32_t> d2s_h;
   int64_t V = 0;
   ~LrtMaps() {
-    if (ctx && dense2slot) mgx_ctx_free(ctx, dense2slot);
+    if (dense2slot) (void)hipFree(dense2slot);
   }
 };
 
 mgx_status lrt_build_maps(mgx_context *ctx, int64_t V, const int64_t *dense_to_mg,
                           LrtMaps *m) {
-  m->ctx = ctx;
   m->V = V;
   m->d2s_h.assign(V > 0 ? V : 1, -1);
   for (int64_t v = 0; v < V; ++v) m->d2s_h[v] = lrt_slot(dense_to_mg[v]);
   MGX_TRY(lrt_grow_slots(ctx, (int64_t)g_l.slot2mg.size()));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m->dense2slot, (V > 0 ? V : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m->dense2slot, m->d2s_h.data(), (V > 0 ? V : 1) * 4,
                              hipMemcpyHostToDevice, ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
@@ -499,8 +497,8 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
   if (n == 0) return MGX_OK;
   int32_t *d_nodes = nullptr;
   uint32_t *d_counts = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_nodes, n * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_counts, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_nodes, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_counts, n * 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_nodes, nodes_dense.data(), n * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   hipLaunchKernelGGL(k_set_count, dim3((uint32_t)grid_for(n)), dim3(kBlock), 0,
@@ -515,8 +513,8 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
     while (cap < need) cap *= 2;
     int32_t *nl = nullptr;
     double *np = nullptr;
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nl, cap * 4));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &np, cap * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&nl, cap * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&np, cap * 8));
     if (g_l.pool_used > 0) {
       MGX_HIP_TRY(hipMemcpyAsync(nl, g_l.lab, g_l.pool_used * 4,
                                  hipMemcpyDeviceToDevice, ctx->stream));
@@ -524,15 +522,15 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
                                  ctx->stream));
     }
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-    if (g_l.lab) mgx_ctx_free(ctx, g_l.lab);
-    if (g_l.p) mgx_ctx_free(ctx, g_l.p);
+    if (g_l.lab) (void)hipFree(g_l.lab);
+    if (g_l.p) (void)hipFree(g_l.p);
     g_l.lab = nl;
     g_l.p = np;
     g_l.pool_cap = cap;
   }
   for (auto &o : seg_off) o += (uint64_t)g_l.pool_used;
   uint64_t *d_seg = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_seg, n * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_seg, n * 8));
   MGX_HIP_TRY(hipMemcpyAsync(d_seg, seg_off.data(), n * 8, hipMemcpyHostToDevice,
                              ctx->stream));
   SetArgs A;
@@ -554,9 +552,9 @@ mgx_status lrt_set_structures(mgx_context *ctx, mgx_graph *g, const LrtMaps &m,
                      A);
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   g_l.pool_used += (int64_t)total;
-  mgx_ctx_free(ctx, d_nodes);
-  mgx_ctx_free(ctx, d_counts);
-  mgx_ctx_free(ctx, d_seg);
+  (void)hipFree(d_nodes);
+  (void)hipFree(d_counts);
+  (void)hipFree(d_seg);
   return MGX_OK;
 }
 
@@ -574,9 +572,9 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   int32_t *d_nodes = nullptr;
   uint8_t *d_sel = nullptr;
   uint32_t *d_bound = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_nodes, n * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_sel, n));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_bound, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_nodes, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_sel, n));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_bound, n * 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_nodes, candidates.data(), n * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   hipLaunchKernelGGL(k_distinct, dim3((uint32_t)grid_for(n)), dim3(kBlock), 0, ctx->stream,
@@ -605,10 +603,10 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   double *scr_p = nullptr;
   uint64_t *d_scr_off = nullptr;
   uint32_t *d_cnt = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &scr_lab, (scr_total > 0 ? scr_total : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &scr_p, (scr_total > 0 ? scr_total : 1) * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_scr_off, n * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_cnt, n * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&scr_lab, (scr_total > 0 ? scr_total : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&scr_p, (scr_total > 0 ? scr_total : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_scr_off, n * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_cnt, n * 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_scr_off, scr_off.data(), n * 8, hipMemcpyHostToDevice,
                              ctx->stream));
   PropArgs P;
@@ -650,13 +648,13 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   }
   *none_updated = !any;
   if (!any) {
-    mgx_ctx_free(ctx, d_nodes);
-    mgx_ctx_free(ctx, d_sel);
-    mgx_ctx_free(ctx, d_bound);
-    mgx_ctx_free(ctx, scr_lab);
-    mgx_ctx_free(ctx, scr_p);
-    mgx_ctx_free(ctx, d_scr_off);
-    mgx_ctx_free(ctx, d_cnt);
+    (void)hipFree(d_nodes);
+    (void)hipFree(d_sel);
+    (void)hipFree(d_bound);
+    (void)hipFree(scr_lab);
+    (void)hipFree(scr_p);
+    (void)hipFree(d_scr_off);
+    (void)hipFree(d_cnt);
     return MGX_OK;
   }
 
@@ -679,14 +677,14 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   uint32_t *d_out_len = nullptr;
   unsigned long long *d_most = nullptr;
   uint32_t *d_any = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nlab, (total > 0 ? total : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &np, (total > 0 ? total : 1) * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_new_off, n_slots * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_slot_sel, n_slots * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_out_off, n_slots * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_out_len, n_slots * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_most, 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_any, 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nlab, (total > 0 ? total : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&np, (total > 0 ? total : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_new_off, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_slot_sel, n_slots * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_out_off, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_out_len, n_slots * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_most, 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_any, 4));
   MGX_HIP_TRY(hipMemcpyAsync(d_new_off, new_off.data(), n_slots * 8,
                              hipMemcpyHostToDevice, ctx->stream));
   MGX_HIP_TRY(hipMemcpyAsync(d_slot_sel, slot_sel.data(), n_slots * 4,
@@ -722,27 +720,27 @@ mgx_status lrt_iteration(mgx_context *ctx, const LrtMaps &m, const uint32_t *row
   MGX_HIP_TRY(hipMemcpyAsync(g_l.len, d_out_len, n_slots * 4, hipMemcpyDeviceToDevice,
                              ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  mgx_ctx_free(ctx, g_l.lab);
-  mgx_ctx_free(ctx, g_l.p);
+  (void)hipFree(g_l.lab);
+  (void)hipFree(g_l.p);
   g_l.lab = nlab;
   g_l.p = np;
   g_l.pool_used = (int64_t)total;
   g_l.pool_cap = (int64_t)(total > 0 ? total : 1);
   *most_updates = most;
 
-  mgx_ctx_free(ctx, d_nodes);
-  mgx_ctx_free(ctx, d_sel);
-  mgx_ctx_free(ctx, d_bound);
-  mgx_ctx_free(ctx, scr_lab);
-  mgx_ctx_free(ctx, scr_p);
-  mgx_ctx_free(ctx, d_scr_off);
-  mgx_ctx_free(ctx, d_cnt);
-  mgx_ctx_free(ctx, d_new_off);
-  mgx_ctx_free(ctx, d_slot_sel);
-  mgx_ctx_free(ctx, d_out_off);
-  mgx_ctx_free(ctx, d_out_len);
-  mgx_ctx_free(ctx, d_most);
-  mgx_ctx_free(ctx, d_any);
+  (void)hipFree(d_nodes);
+  (void)hipFree(d_sel);
+  (void)hipFree(d_bound);
+  (void)hipFree(scr_lab);
+  (void)hipFree(scr_p);
+  (void)hipFree(d_scr_off);
+  (void)hipFree(d_cnt);
+  (void)hipFree(d_new_off);
+  (void)hipFree(d_slot_sel);
+  (void)hipFree(d_out_off);
+  (void)hipFree(d_out_len);
+  (void)hipFree(d_most);
+  (void)hipFree(d_any);
   return MGX_OK;
 }
 
@@ -751,8 +749,8 @@ mgx_status lrt_all_labels(mgx_context *ctx, const LrtMaps &m, int64_t *out_label
   const int64_t n_slots = (int64_t)g_l.slot2mg.size();
   if (n_slots == 0 || m.V == 0) return MGX_OK;
   int64_t *d_s2mg = nullptr, *d_raw = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_s2mg, n_slots * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_raw, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_s2mg, n_slots * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_raw, n_slots * 8));
   MGX_HIP_TRY(hipMemcpyAsync(d_s2mg, g_l.slot2mg.data(), n_slots * 8,
                              hipMemcpyHostToDevice, ctx->stream));
   hipLaunchKernelGGL(k_argmax, dim3((uint32_t)grid_for(n_slots)), dim3(kBlock), 0,
@@ -762,8 +760,8 @@ mgx_status lrt_all_labels(mgx_context *ctx, const LrtMaps &m, int64_t *out_label
   MGX_HIP_TRY(hipMemcpyAsync(raw.data(), d_raw, n_slots * 8, hipMemcpyDeviceToHost,
                              ctx->stream));
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  mgx_ctx_free(ctx, d_s2mg);
-  mgx_ctx_free(ctx, d_raw);
+  (void)hipFree(d_s2mg);
+  (void)hipFree(d_raw);
   // renumber in ascending label-mg-id order (:135-141); only ALIVE slots
   std::set<int64_t> ordered;
   for (int64_t s = 0; s < n_slots; ++s)
@@ -845,7 +843,7 @@ extern "C" int mgx_lrt_initialized(void) { return g_l.calculated ? 1 : 0; }
 
 extern "C" mgx_status mgx_lrt_reset(mgx_context *ctx) {
   (void)ctx;
-  lrt_free(ctx);
+  lrt_free();
   return MGX_OK;
 }
 
@@ -871,7 +869,7 @@ extern "C" mgx_status mgx_lrt_set(mgx_context *ctx, mgx_graph *g,
                                   int64_t *out_label) {
   MGX_HIP_TRY(hipSetDevice(ctx->device));
   MGX_TRY(lrt_check_graph(g, directed != 0));
-  lrt_free(ctx);
+  lrt_free();
   g_l.directed = directed != 0;
   g_l.weighted = weighted != 0;
   g_l.sim_th = similarity_threshold;

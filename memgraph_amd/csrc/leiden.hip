@@ -87,11 +87,11 @@ struct Level {
   int64_t ne2 = 0;
 };
 
-void free_level(mgx_context *ctx, Level *L) {
-  if (L->row_ptr) mgx_ctx_free(ctx, L->row_ptr);
-  if (L->col) mgx_ctx_free(ctx, L->col);
-  if (L->w) mgx_ctx_free(ctx, L->w);
-  if (L->node_w) mgx_ctx_free(ctx, L->node_w);
+void free_level(Level *L) {
+  if (L->row_ptr) (void)hipFree(L->row_ptr);
+  if (L->col) (void)hipFree(L->col);
+  if (L->w) (void)hipFree(L->w);
+  if (L->node_w) (void)hipFree(L->node_w);
   *L = Level{};
 }
 
@@ -539,10 +539,10 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
   Level L;
   L.nv = V;
   L.ne2 = (int64_t)col2.size();
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.row_ptr, (V + 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.col, (L.ne2 > 0 ? L.ne2 : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.w, (L.ne2 > 0 ? L.ne2 : 1) * 8));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.node_w, V * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.row_ptr, (V + 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.col, (L.ne2 > 0 ? L.ne2 : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.w, (L.ne2 > 0 ? L.ne2 : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&L.node_w, V * 8));
   MGX_HIP_TRY(hipMemcpyAsync(L.row_ptr, rp2.data(), (V + 1) * 4, hipMemcpyHostToDevice,
                              ctx->stream));
   if (L.ne2 > 0) {
@@ -568,8 +568,8 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
   // it; at level 0 it is singleton.
   int32_t *d_comm = nullptr;
   uint32_t *d_csize = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_comm, V * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_csize, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_comm, V * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_csize, V * 4));
   hipLaunchKernelGGL(k_iota32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
                      V, d_comm);
   hipLaunchKernelGGL(k_fillu32, dim3((uint32_t)grid_for(V)), dim3(kBlock), 0, ctx->stream,
@@ -609,13 +609,13 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     int32_t *d_pkeys = nullptr;
     double *d_pvals = nullptr;
     unsigned long long *d_cnt = nullptr;
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_srows, (srows.empty() ? 1 : srows.size()) * 4));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_brows, (brows.empty() ? 1 : brows.size()) * 4));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_target, nv * 4));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_poff, caps.size() * 8));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_pkeys, (pool_total ? pool_total : 1) * 4));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_pvals, (pool_total ? pool_total : 1) * 8));
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_cnt, 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_srows, (srows.empty() ? 1 : srows.size()) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_brows, (brows.empty() ? 1 : brows.size()) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_target, nv * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_poff, caps.size() * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_pkeys, (pool_total ? pool_total : 1) * 4));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_pvals, (pool_total ? pool_total : 1) * 8));
+    MGX_HIP_TRY(mgx_hip_malloc(&d_cnt, 8));
     if (!srows.empty())
       MGX_HIP_TRY(hipMemcpyAsync(d_srows, srows.data(), srows.size() * 4,
                                  hipMemcpyHostToDevice, ctx->stream));
@@ -628,7 +628,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     unsigned long long empties_before = 0;
     {
       unsigned long long *d_t = nullptr;
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_t, 16));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_t, 16));
       MGX_HIP_TRY(hipMemsetAsync(d_t, 0, 16, ctx->stream));
       hipLaunchKernelGGL(k_count_nonsingleton, dim3((uint32_t)grid_for(nv)), dim3(kBlock),
                          0, ctx->stream, nv, d_csize, d_t, d_t + 1);
@@ -636,7 +636,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       MGX_HIP_TRY(hipMemcpyAsync(h_t, d_t, 16, hipMemcpyDeviceToHost, ctx->stream));
       MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
       empties_before = (unsigned long long)nv - h_t[1];
-      mgx_ctx_free(ctx, d_t);
+      (void)hipFree(d_t);
     }
 
     MoveArgs M;
@@ -697,7 +697,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
     unsigned long long nonsingleton = 0, nonempty = 0;
     {
       unsigned long long *d_t = nullptr;
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_t, 16));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_t, 16));
       MGX_HIP_TRY(hipMemsetAsync(d_t, 0, 16, ctx->stream));
       hipLaunchKernelGGL(k_count_nonsingleton, dim3((uint32_t)grid_for(nv)), dim3(kBlock),
                          0, ctx->stream, nv, d_csize, d_t, d_t + 1);
@@ -706,7 +706,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
       nonsingleton = h_t[0];
       nonempty = h_t[1];
-      mgx_ctx_free(ctx, d_t);
+      (void)hipFree(d_t);
     }
     const unsigned long long move_empties =
         ((unsigned long long)nv - nonempty) - empties_before;
@@ -738,16 +738,16 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       uint32_t *d_memoff = nullptr, *d_rsize = nullptr;
       double *d_extw = nullptr, *d_edgew = nullptr, *d_prob = nullptr;
       int32_t *d_nbc = nullptr, *d_cslot = nullptr;
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_members, nv * 4));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_locidx, nv * 4));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_rcomm, nv * 4));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_memoff, (nv + 1) * 4));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_rsize, nv * 4));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_extw, nv * 8));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_edgew, nv * 8));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_prob, nv * 8));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_nbc, nv * 4));
-      MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_cslot, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_members, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_locidx, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_rcomm, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_memoff, (nv + 1) * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_rsize, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_extw, nv * 8));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_edgew, nv * 8));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_prob, nv * 8));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_nbc, nv * 4));
+      MGX_HIP_TRY(mgx_hip_malloc(&d_cslot, nv * 4));
       MGX_HIP_TRY(hipMemcpyAsync(d_members, members.data(), nv * 4,
                                  hipMemcpyHostToDevice, ctx->stream));
       MGX_HIP_TRY(hipMemcpyAsync(d_memoff, mem_off.data(), (nv + 1) * 4,
@@ -787,16 +787,16 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
       MGX_HIP_TRY(hipMemcpyAsync(rcomm_h.data(), d_rcomm, nv * 4, hipMemcpyDeviceToHost,
                                  ctx->stream));
       MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-      mgx_ctx_free(ctx, d_members);
-      mgx_ctx_free(ctx, d_locidx);
-      mgx_ctx_free(ctx, d_memoff);
-      mgx_ctx_free(ctx, d_rsize);
-      mgx_ctx_free(ctx, d_extw);
-      mgx_ctx_free(ctx, d_edgew);
-      mgx_ctx_free(ctx, d_prob);
-      mgx_ctx_free(ctx, d_nbc);
-      mgx_ctx_free(ctx, d_cslot);
-      mgx_ctx_free(ctx, d_rcomm);
+      (void)hipFree(d_members);
+      (void)hipFree(d_locidx);
+      (void)hipFree(d_memoff);
+      (void)hipFree(d_rsize);
+      (void)hipFree(d_extw);
+      (void)hipFree(d_edgew);
+      (void)hipFree(d_prob);
+      (void)hipFree(d_nbc);
+      (void)hipFree(d_cslot);
+      (void)hipFree(d_rcomm);
 
       // global refined community id per node: the member slot of its local
       // refined community (== another node's global member position)
@@ -901,7 +901,7 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
         }
 
         // upload the new level
-        free_level(ctx, &L);
+        free_level(&L);
         L.nv = n_new;
         std::vector<uint32_t> rp_up(n_new + 1, 0);
         std::vector<int32_t> col_up;
@@ -914,10 +914,10 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
           rp_up[i2 + 1] = (uint32_t)col_up.size();
         }
         L.ne2 = (int64_t)col_up.size();
-        MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.row_ptr, (n_new + 1) * 4));
-        MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.col, (L.ne2 ? L.ne2 : 1) * 4));
-        MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.w, (L.ne2 ? L.ne2 : 1) * 8));
-        MGX_HIP_TRY(mgx_ctx_malloc(ctx, &L.node_w, n_new * 8));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.row_ptr, (n_new + 1) * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.col, (L.ne2 ? L.ne2 : 1) * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.w, (L.ne2 ? L.ne2 : 1) * 8));
+        MGX_HIP_TRY(mgx_hip_malloc(&L.node_w, n_new * 8));
         MGX_HIP_TRY(hipMemcpyAsync(L.row_ptr, rp_up.data(), (n_new + 1) * 4,
                                    hipMemcpyHostToDevice, ctx->stream));
         if (L.ne2) {
@@ -928,10 +928,10 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
         }
         MGX_HIP_TRY(hipMemcpyAsync(L.node_w, nw_new.data(), n_new * 8,
                                    hipMemcpyHostToDevice, ctx->stream));
-        mgx_ctx_free(ctx, d_comm);
-        mgx_ctx_free(ctx, d_csize);
-        MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_comm, n_new * 4));
-        MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_csize, n_new * 4));
+        (void)hipFree(d_comm);
+        (void)hipFree(d_csize);
+        MGX_HIP_TRY(mgx_hip_malloc(&d_comm, n_new * 4));
+        MGX_HIP_TRY(mgx_hip_malloc(&d_csize, n_new * 4));
         MGX_HIP_TRY(hipMemcpyAsync(d_comm, comm_new.data(), n_new * 4,
                                    hipMemcpyHostToDevice, ctx->stream));
         std::vector<uint32_t> cs_new((size_t)n_new, 0);
@@ -945,18 +945,18 @@ extern "C" mgx_status mgx_leiden(mgx_context *ctx, mgx_graph *g, double gamma,
         ++level;
       }
     }
-    mgx_ctx_free(ctx, d_srows);
-    mgx_ctx_free(ctx, d_brows);
-    mgx_ctx_free(ctx, d_target);
-    mgx_ctx_free(ctx, d_poff);
-    mgx_ctx_free(ctx, d_pkeys);
-    mgx_ctx_free(ctx, d_pvals);
-    mgx_ctx_free(ctx, d_cnt);
+    (void)hipFree(d_srows);
+    (void)hipFree(d_brows);
+    (void)hipFree(d_target);
+    (void)hipFree(d_poff);
+    (void)hipFree(d_pkeys);
+    (void)hipFree(d_pvals);
+    (void)hipFree(d_cnt);
     if (iters >= max_iterations) done = true;
   }
-  mgx_ctx_free(ctx, d_comm);
-  mgx_ctx_free(ctx, d_csize);
-  free_level(ctx, &L);
+  (void)hipFree(d_comm);
+  (void)hipFree(d_csize);
+  free_level(&L);
   if (st != MGX_OK) return st;
 
   // compose per-node hierarchies from the dendrogram levels

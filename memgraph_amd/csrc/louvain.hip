@@ -1267,7 +1267,7 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
 
   out->nv = n_clusters;
   ctx->ensure_margin((size_t)(n_clusters + 1) * 4);
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &out->row_ptr, (n_clusters + 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&out->row_ptr, (n_clusters + 1) * 4));
   if (louvain_trace()) {
     // Pre-scan probe: counts must still hold the pair histogram, and the
     // freshly hipMalloc'd row_ptr must not alias the pool-allocated counts
@@ -1330,8 +1330,8 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
     return MGX_ERR_HIP;
   }
   ctx->ensure_margin((size_t)(total > 0 ? total : 1) * 12);
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &out->col, (total > 0 ? total : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &out->w, (total > 0 ? total : 1) * 8));
+  MGX_HIP_TRY(mgx_hip_malloc(&out->col, (total > 0 ? total : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&out->w, (total > 0 ? total : 1) * 8));
   if (louvain_trace()) {
     // Allocator-overlap probe: the async-pool DevBufs vs the plain hipMalloc
     // level buffers must be disjoint VA ranges.
@@ -1386,11 +1386,11 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   return MGX_OK;
 }
 
-void free_level(mgx_context *ctx, Level *L, bool own) {
+void free_level(Level *L, bool own) {
   if (!own) return;
-  if (L->row_ptr) mgx_ctx_free(ctx, L->row_ptr);
-  if (L->col) mgx_ctx_free(ctx, L->col);
-  if (L->w) mgx_ctx_free(ctx, L->w);
+  if (L->row_ptr) (void)hipFree(L->row_ptr);
+  if (L->col) (void)hipFree(L->col);
+  if (L->w) (void)hipFree(L->w);
   L->row_ptr = nullptr;
   L->col = nullptr;
   L->w = nullptr;
@@ -1487,7 +1487,7 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
       MGX_LDBG(ctx, "coarsen from nv=%lld to %lld", (long long)L.nv,
                (long long)n_clusters);
       status = coarsen(ctx, L, C.as<int32_t>(), n_clusters, &next);
-      free_level(ctx, &L, own_level);
+      free_level(&L, own_level);
       if (status != MGX_OK) break;
       L = next;
       own_level = true;
@@ -1496,7 +1496,7 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
       break;
     }
   }
-  free_level(ctx, &L, own_level);
+  free_level(&L, own_level);
   if (status != MGX_OK) return status;
 
   // Download.

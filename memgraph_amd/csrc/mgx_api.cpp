@@ -98,11 +98,6 @@ mgx_status mgx_context::alloc_async(void **p, size_t bytes) {
   return MGX_OK;
 }
 
-mgx_status mgx_ctx_alloc_raw(mgx_context *c, void **p, size_t bytes) {
-  return c->alloc_async(p, bytes);
-}
-mgx_status mgx_ctx_free_raw(mgx_context *c, void *p) { return c->free_async(p); }
-
 mgx_status mgx_context::free_async(void *p) {
   if (!p) return MGX_OK;
   if (no_alloc_cache()) {
@@ -245,9 +240,9 @@ extern "C" mgx_status mgx_graph_from_coo(mgx_context *ctx, const int64_t *src,
   auto *g = new mgx_graph();
   mgx_status s = mgx_build_from_device_coo(ctx, d_src, d_dst, d_w, n_vertices, n_edges,
                                            flags, g);
-  mgx_ctx_free(ctx, d_src);
-  mgx_ctx_free(ctx, d_dst);
-  if (d_w) mgx_ctx_free(ctx, d_w);
+  (void)hipFree(d_src);
+  (void)hipFree(d_dst);
+  if (d_w) (void)hipFree(d_w);
   if (s != MGX_OK) {
     (void)mgx_graph_destroy(ctx, g);
     return s;
@@ -325,8 +320,8 @@ extern "C" mgx_status mgx_graph_rmat_sharded(mgx_context *ctx, int scale, int64_
   }
   MGX_HIP_TRY(hipSetDevice(ctx->device));
   int32_t *d_src = nullptr, *d_dst = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_src, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
+  MGX_HIP_TRY(mgx_hip_malloc(&d_dst, (n_edges > 0 ? n_edges : 1) * sizeof(int32_t)));
   mgx_status s = mgx_gen_rmat_device(ctx, scale, n_edges, seed, a, b, c, d_src, d_dst);
   mgx_graph *g = nullptr;
   if (s == MGX_OK) {
@@ -337,8 +332,8 @@ extern "C" mgx_status mgx_graph_rmat_sharded(mgx_context *ctx, int scale, int64_
       g->row_end = row_end;  // keep the padded shard size for allgather
     }
   }
-  mgx_ctx_free(ctx, d_src);
-  mgx_ctx_free(ctx, d_dst);
+  (void)hipFree(d_src);
+  (void)hipFree(d_dst);
   if (s != MGX_OK) {
     if (g) (void)mgx_graph_destroy(ctx, g);
     return s;
@@ -349,23 +344,24 @@ extern "C" mgx_status mgx_graph_rmat_sharded(mgx_context *ctx, int scale, int64_
 
 extern "C" mgx_status mgx_graph_destroy(mgx_context *ctx, mgx_graph *g) {
   if (!g) return MGX_OK;
-  if (g->in_row_ptr) mgx_ctx_free(ctx, g->in_row_ptr);
-  if (g->in_col) mgx_ctx_free(ctx, g->in_col);
-  if (g->out_degree) mgx_ctx_free(ctx, g->out_degree);
-  if (g->inv_outdeg) mgx_ctx_free(ctx, g->inv_outdeg);
-  if (g->out_row_ptr) mgx_ctx_free(ctx, g->out_row_ptr);
-  if (g->out_col) mgx_ctx_free(ctx, g->out_col);
-  if (g->in_w) mgx_ctx_free(ctx, g->in_w);
-  if (g->bins_out.rows) mgx_ctx_free(ctx, g->bins_out.rows);
-  if (g->sym_row_ptr) mgx_ctx_free(ctx, g->sym_row_ptr);
-  if (g->sym_col) mgx_ctx_free(ctx, g->sym_col);
-  if (g->sym_w) mgx_ctx_free(ctx, g->sym_w);
-  if (g->bins_in.rows) mgx_ctx_free(ctx, g->bins_in.rows);
-  if (g->bins_sym.rows) mgx_ctx_free(ctx, g->bins_sym.rows);
-  if (g->stripe_ptr) mgx_ctx_free(ctx, g->stripe_ptr);
-  if (g->order) mgx_ctx_free(ctx, g->order);
+  (void)ctx;
+  if (g->in_row_ptr) (void)hipFree(g->in_row_ptr);
+  if (g->in_col) (void)hipFree(g->in_col);
+  if (g->out_degree) (void)hipFree(g->out_degree);
+  if (g->inv_outdeg) (void)hipFree(g->inv_outdeg);
+  if (g->out_row_ptr) (void)hipFree(g->out_row_ptr);
+  if (g->out_col) (void)hipFree(g->out_col);
+  if (g->in_w) (void)hipFree(g->in_w);
+  if (g->bins_out.rows) (void)hipFree(g->bins_out.rows);
+  if (g->sym_row_ptr) (void)hipFree(g->sym_row_ptr);
+  if (g->sym_col) (void)hipFree(g->sym_col);
+  if (g->sym_w) (void)hipFree(g->sym_w);
+  if (g->bins_in.rows) (void)hipFree(g->bins_in.rows);
+  if (g->bins_sym.rows) (void)hipFree(g->bins_sym.rows);
+  if (g->stripe_ptr) (void)hipFree(g->stripe_ptr);
+  if (g->order) (void)hipFree(g->order);
   for (int i = 0; i < 16; ++i) {
-    if (g->stripe_bins[i].rows) mgx_ctx_free(ctx, g->stripe_bins[i].rows);
+    if (g->stripe_bins[i].rows) (void)hipFree(g->stripe_bins[i].rows);
   }
   delete g;
   return MGX_OK;

@@ -55,24 +55,6 @@ inline hipError_t mgx_hip_malloc(T **p, size_t bytes) {
   return hipMalloc((void **)p, mgx_safe_size(bytes));
 }
 
-// The alignment workaround above is insufficient once the VA space has
-// holes: the runtime's hole fit-check compares only (size mod 2^32), so
-// with an exact multiple ANY hole "fits" (measured: an 8.59 GB request
-// landed in a 525 MB hole, overlapping five live buffers). The robust
-// dodge is to never return VA to the runtime while the process lives:
-// every allocation AND free in this library goes through the context's
-// caching free list (mgx_ctx_malloc/mgx_ctx_free), so no holes form.
-struct mgx_context;
-mgx_status mgx_ctx_alloc_raw(mgx_context *c, void **p, size_t bytes);
-mgx_status mgx_ctx_free_raw(mgx_context *c, void *p);
-
-template <typename T>
-inline hipError_t mgx_ctx_malloc(mgx_context *c, T **p, size_t bytes) {
-  return mgx_ctx_alloc_raw(c, (void **)p, bytes) == MGX_OK ? hipSuccess
-                                                           : hipErrorOutOfMemory;
-}
-inline void mgx_ctx_free(mgx_context *c, void *p) { (void)mgx_ctx_free_raw(c, p); }
-
 struct mgx_context {
   int device = -1;
   hipStream_t stream = nullptr;

@@ -257,9 +257,9 @@ mgx_status ensure_pool(mgx_context *ctx, int64_t need_cap) {
   while (cap < need_cap) cap *= 2;
   uint32_t *nw = nullptr, *np = nullptr;
   int32_t *nn = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nw, cap * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &np, cap * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nn, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nw, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&np, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nn, cap * 4));
   if (g_st.pool_used > 0) {
     MGX_HIP_TRY(hipMemcpyAsync(nw, g_st.e_walk, g_st.pool_used * 4,
                                hipMemcpyDeviceToDevice, ctx->stream));
@@ -269,9 +269,9 @@ mgx_status ensure_pool(mgx_context *ctx, int64_t need_cap) {
                                hipMemcpyDeviceToDevice, ctx->stream));
     MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
   }
-  if (g_st.e_walk) mgx_ctx_free(ctx, g_st.e_walk);
-  if (g_st.e_pos) mgx_ctx_free(ctx, g_st.e_pos);
-  if (g_st.e_node) mgx_ctx_free(ctx, g_st.e_node);
+  if (g_st.e_walk) (void)hipFree(g_st.e_walk);
+  if (g_st.e_pos) (void)hipFree(g_st.e_pos);
+  if (g_st.e_node) (void)hipFree(g_st.e_node);
   g_st.e_walk = nw;
   g_st.e_pos = np;
   g_st.e_node = nn;
@@ -286,9 +286,9 @@ mgx_status ensure_walks(mgx_context *ctx, int64_t need) {
   int32_t *ns = nullptr;
   uint32_t *ng = nullptr;
   uint8_t *nd = nullptr;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &ns, cap * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &ng, cap * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &nd, cap));
+  MGX_HIP_TRY(mgx_hip_malloc(&ns, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&ng, cap * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&nd, cap));
   MGX_HIP_TRY(hipMemsetAsync(ng, 0, cap * 4, ctx->stream));
   MGX_HIP_TRY(hipMemsetAsync(nd, 0, cap, ctx->stream));
   if (g_st.n_walks > 0) {
@@ -300,9 +300,9 @@ mgx_status ensure_walks(mgx_context *ctx, int64_t need) {
                                hipMemcpyDeviceToDevice, ctx->stream));
   }
   MGX_HIP_TRY(hipStreamSynchronize(ctx->stream));
-  if (g_st.w_start) mgx_ctx_free(ctx, g_st.w_start);
-  if (g_st.w_gen) mgx_ctx_free(ctx, g_st.w_gen);
-  if (g_st.w_dead) mgx_ctx_free(ctx, g_st.w_dead);
+  if (g_st.w_start) (void)hipFree(g_st.w_start);
+  if (g_st.w_gen) (void)hipFree(g_st.w_gen);
+  if (g_st.w_dead) (void)hipFree(g_st.w_dead);
   g_st.w_start = ns;
   g_st.w_gen = ng;
   g_st.w_dead = nd;
@@ -313,10 +313,10 @@ mgx_status ensure_walks(mgx_context *ctx, int64_t need) {
 mgx_status sync_slot_alive(mgx_context *ctx) {
   const int64_t n = (int64_t)g_st.slot_alive_h.size();
   if (n > g_st.slots_cap_dev) {
-    if (g_st.d_slot_alive) mgx_ctx_free(ctx, g_st.d_slot_alive);
+    if (g_st.d_slot_alive) (void)hipFree(g_st.d_slot_alive);
     int64_t cap = 64;
     while (cap < n) cap *= 2;
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g_st.d_slot_alive, cap));
+    MGX_HIP_TRY(mgx_hip_malloc(&g_st.d_slot_alive, cap));
     g_st.slots_cap_dev = cap;
   }
   if (n > 0) {
@@ -326,15 +326,15 @@ mgx_status sync_slot_alive(mgx_context *ctx) {
   return MGX_OK;
 }
 
-void free_state(mgx_context *ctx) {
-  if (g_st.e_walk) mgx_ctx_free(ctx, g_st.e_walk);
-  if (g_st.e_pos) mgx_ctx_free(ctx, g_st.e_pos);
-  if (g_st.e_node) mgx_ctx_free(ctx, g_st.e_node);
-  if (g_st.w_start) mgx_ctx_free(ctx, g_st.w_start);
-  if (g_st.w_gen) mgx_ctx_free(ctx, g_st.w_gen);
-  if (g_st.w_dead) mgx_ctx_free(ctx, g_st.w_dead);
-  if (g_st.d_cursor) mgx_ctx_free(ctx, g_st.d_cursor);
-  if (g_st.d_slot_alive) mgx_ctx_free(ctx, g_st.d_slot_alive);
+void free_state() {
+  if (g_st.e_walk) (void)hipFree(g_st.e_walk);
+  if (g_st.e_pos) (void)hipFree(g_st.e_pos);
+  if (g_st.e_node) (void)hipFree(g_st.e_node);
+  if (g_st.w_start) (void)hipFree(g_st.w_start);
+  if (g_st.w_gen) (void)hipFree(g_st.w_gen);
+  if (g_st.w_dead) (void)hipFree(g_st.w_dead);
+  if (g_st.d_cursor) (void)hipFree(g_st.d_cursor);
+  if (g_st.d_slot_alive) (void)hipFree(g_st.d_slot_alive);
   g_st = PrOnlineState{};
 }
 
@@ -355,8 +355,8 @@ struct Maps {
   int64_t V = 0;
   mgx_context *ctx = nullptr;
   ~Maps() {
-    if (slot2dense) mgx_ctx_free(ctx, slot2dense);
-    if (dense2slot) mgx_ctx_free(ctx, dense2slot);
+    if (slot2dense) (void)hipFree(slot2dense);
+    if (dense2slot) (void)hipFree(dense2slot);
   }
 };
 
@@ -385,8 +385,8 @@ mgx_status build_maps(mgx_context *ctx, const int64_t *dense_to_mg, int64_t V, b
   std::vector<int32_t> s2d(n_slots > 0 ? n_slots : 1, -1);
   for (int64_t v = 0; v < V; ++v)
     if (d2s[v] >= 0) s2d[d2s[v]] = (int32_t)v;
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m->dense2slot, (V > 0 ? V : 1) * 4));
-  MGX_HIP_TRY(mgx_ctx_malloc(ctx, &m->slot2dense, (n_slots > 0 ? n_slots : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->dense2slot, (V > 0 ? V : 1) * 4));
+  MGX_HIP_TRY(mgx_hip_malloc(&m->slot2dense, (n_slots > 0 ? n_slots : 1) * 4));
   MGX_HIP_TRY(hipMemcpyAsync(m->dense2slot, d2s.data(), (V > 0 ? V : 1) * 4,
                              hipMemcpyHostToDevice, ctx->stream));
   MGX_HIP_TRY(hipMemcpyAsync(m->slot2dense, s2d.data(), (n_slots > 0 ? n_slots : 1) * 4,
@@ -404,7 +404,7 @@ mgx_status run_gen(mgx_context *ctx, const Maps &m, mgx_graph *g,
                    int64_t expected_append) {
   if (n_items == 0) return MGX_OK;
   if (!g_st.d_cursor) {
-    MGX_HIP_TRY(mgx_ctx_malloc(ctx, &g_st.d_cursor, 16));
+    MGX_HIP_TRY(mgx_hip_malloc(&g_st.d_cursor, 16));
   }
   uint64_t eps_bits;
   if (eps >= 1.0) {
@@ -563,8 +563,8 @@ mgx_status rewire_from(mgx_context *ctx, const Maps &m, mgx_graph *g, int32_t fr
 extern "C" int mgx_pronline_initialized(void) { return g_st.initialized ? 1 : 0; }
 
 extern "C" mgx_status mgx_pronline_reset(mgx_context *ctx) {
-
-  free_state(ctx);
+  (void)ctx;
+  free_state();
   return MGX_OK;
 }
 
@@ -576,7 +576,7 @@ extern "C" mgx_status mgx_pronline_set(mgx_context *ctx, mgx_graph *g,
     return MGX_ERR_INVALID_ARGUMENT;
   }
   MGX_HIP_TRY(hipSetDevice(ctx->device));
-  free_state(ctx);
+  free_state();
   g_st.R = R > 0 ? R : 1;
   g_st.eps = eps;
   g_st.seed_mixed = mgx_seed_mix(seed);
