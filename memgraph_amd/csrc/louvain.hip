@@ -1332,6 +1332,50 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
   ctx->ensure_margin((size_t)(total > 0 ? total : 1) * 12);
   MGX_HIP_TRY(mgx_hip_malloc(&out->col, (total > 0 ? total : 1) * 4));
   MGX_HIP_TRY(mgx_hip_malloc(&out->w, (total > 0 ? total : 1) * 8));
+  {
+    // ALWAYS-ON overlap gate. The box's ROCm 7.0.x runtime mis-places
+    // multi-GiB hipMallocs once the VA space is fragmented: at RMAT-25+
+    // the total-sized col/w come back overlapping five live buffers (the
+    // fit check appears to use only the low 32 bits of the size; an
+    // isolated fresh-process repro allocates the same sizes correctly).
+    // Scattering through such a mapping faults the GPU, so detect the
+    // overlap host-side and fail cleanly instead.
+    struct Rg {
+      const void *p;
+      size_t bytes;
+    } rgs[] = {
+        {counts.p, (size_t)n_clusters * 4},
+        {self_present.p, (size_t)n_clusters * 4},
+        {row_of_entry.p, (size_t)ne2 * 4},
+        {keys.p, (size_t)ne2 * 8},
+        {vals.p, (size_t)ne2 * 8},
+        {keys_sorted.p, (size_t)ne2 * 8},
+        {vals_sorted.p, (size_t)ne2 * 8},
+        {u_keys.p, (size_t)ne2 * 8},
+        {u_vals.p, (size_t)ne2 * 8},
+        {ctx->workspace, ctx->workspace_bytes},
+        {in.row_ptr, (size_t)(in.nv + 1) * 4},
+        {in.col, (size_t)ne2 * 4},
+        {in.w, (size_t)ne2 * 8},
+        {out->row_ptr, (size_t)(n_clusters + 1) * 4},
+        {out->col, (size_t)(total > 0 ? total : 1) * 4},
+        {out->w, (size_t)(total > 0 ? total : 1) * 8},
+    };
+    const size_t nr = sizeof(rgs) / sizeof(rgs[0]);
+    for (size_t a = 0; a < nr; ++a)
+      for (size_t b = a + 1; b < nr; ++b) {
+        const char *a0 = (const char *)rgs[a].p, *a1 = a0 + rgs[a].bytes;
+        const char *b0 = (const char *)rgs[b].p, *b1 = b0 + rgs[b].bytes;
+        if (a0 && b0 && a0 < b1 && b0 < a1) {
+          mgx_set_error(
+              "louvain coarsen: the runtime returned overlapping device "
+              "allocations (ranges %zu and %zu; ncl=%lld total=%u) — known "
+              "ROCm 7.0.x VA-placement fault at this scale, see DESIGN.md",
+              a, b, (long long)n_clusters, total);
+          return MGX_ERR_HIP;
+        }
+      }
+  }
   if (louvain_trace()) {
     // Allocator-overlap probe: the async-pool DevBufs vs the plain hipMalloc
     // level buffers must be disjoint VA ranges.
