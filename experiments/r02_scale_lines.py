@@ -33,8 +33,9 @@ def bench(name, flags, fn):
 ids = np.arange(V, dtype=np.int64)
 
 if "leiden" in algos:
+    LEIDEN_ITERS = int(os.environ.get("MGX_LINES_LEIDEN_ITERS", str(1 << 62)))
     def run_leiden(g):
-        hier, levels = n.leiden(ctx, g, V, seed=42)
+        hier, levels = n.leiden(ctx, g, V, seed=42, max_iterations=LEIDEN_ITERS)
         top = hier[np.arange(V), levels - 1]
         return f"levels max={levels.max()}, top communities={len(np.unique(top))}"
     bench(f"leiden RMAT-{SCALE}w", BUILD_SYM_CSR | BUILD_WEIGHTED, run_leiden)
