@@ -106,6 +106,7 @@ struct Level {
   int32_t *col = nullptr;       // [ne2]
   double *w = nullptr;          // [ne2]
   int64_t ne2 = 0;              // entries (each input edge twice)
+  bool arena = false;           // col/w live in ctx->louv_* (not freed here)
 };
 
 // ---------- generic small kernels ----------------------------------------
@@ -1329,9 +1330,22 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
         total, (long long)n_clusters, (long long)n_pairs);
     return MGX_ERR_HIP;
   }
-  ctx->ensure_margin((size_t)(total > 0 ? total : 1) * 12);
-  MGX_HIP_TRY(mgx_hip_malloc(&out->col, (total > 0 ? total : 1) * 4));
-  MGX_HIP_TRY(mgx_hip_malloc(&out->w, (total > 0 ? total : 1) * 8));
+  // col/w come from the entry-allocated ping-pong arenas when they fit
+  // (they always do: total <= ne2_0 + nv_0); a direct malloc is only the
+  // fallback for graphs whose first level was never arena-sized.
+  const int flip = ctx->louv_flip ^= 1;
+  const size_t need_c = (size_t)(total > 0 ? total : 1) * 4;
+  const size_t need_w = (size_t)(total > 0 ? total : 1) * 8;
+  if (ctx->louv_col[flip] && need_c <= ctx->louv_col_bytes[flip] &&
+      ctx->louv_w[flip] && need_w <= ctx->louv_w_bytes[flip]) {
+    out->col = (int32_t *)ctx->louv_col[flip];
+    out->w = (double *)ctx->louv_w[flip];
+    out->arena = true;
+  } else {
+    ctx->ensure_margin(need_c + need_w);
+    MGX_HIP_TRY(mgx_hip_malloc(&out->col, need_c));
+    MGX_HIP_TRY(mgx_hip_malloc(&out->w, need_w));
+  }
   {
     // ALWAYS-ON overlap gate. The box's ROCm 7.0.x runtime mis-places
     // multi-GiB hipMallocs once the VA space is fragmented: at RMAT-25+
@@ -1433,8 +1447,10 @@ mgx_status coarsen(mgx_context *ctx, const Level &in, const int32_t *C, int64_t 
 void free_level(Level *L, bool own) {
   if (!own) return;
   if (L->row_ptr) (void)hipFree(L->row_ptr);
-  if (L->col) (void)hipFree(L->col);
-  if (L->w) (void)hipFree(L->w);
+  if (!L->arena) {
+    if (L->col) (void)hipFree(L->col);
+    if (L->w) (void)hipFree(L->w);
+  }
   L->row_ptr = nullptr;
   L->col = nullptr;
   L->w = nullptr;
@@ -1475,6 +1491,28 @@ mgx_status mgx_louvain_impl(mgx_context *ctx, mgx_graph *g, double threshold,
   L.row_ptr = g->sym_row_ptr;
   L.col = g->sym_col;
   L.ne2 = 2 * g->n_edges;
+  // Size the coarsen col/w arenas now, before any per-level VA churn: a
+  // coarse level has at most ne2 + nv entries (every off-diagonal pair
+  // twice + one self loop per cluster).
+  {
+    const size_t bound = (size_t)(L.ne2 + nv0 + 1);
+    for (int i = 0; i < 2; ++i) {
+      if (ctx->louv_col_bytes[i] < bound * 4) {
+        if (ctx->louv_col[i]) (void)hipFree(ctx->louv_col[i]);
+        ctx->louv_col[i] = nullptr;
+        ctx->louv_col_bytes[i] = 0;
+        MGX_HIP_TRY(mgx_hip_malloc(&ctx->louv_col[i], bound * 4));
+        ctx->louv_col_bytes[i] = bound * 4;
+      }
+      if (ctx->louv_w_bytes[i] < bound * 8) {
+        if (ctx->louv_w[i]) (void)hipFree(ctx->louv_w[i]);
+        ctx->louv_w[i] = nullptr;
+        ctx->louv_w_bytes[i] = 0;
+        MGX_HIP_TRY(mgx_hip_malloc(&ctx->louv_w[i], bound * 8));
+        ctx->louv_w_bytes[i] = bound * 8;
+      }
+    }
+  }
   DevBuf w0;
   MGX_HIP_TRY(w0.alloc(ctx, L.ne2 * 8));
   if (g->sym_w) {

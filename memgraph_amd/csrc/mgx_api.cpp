@@ -183,6 +183,10 @@ extern "C" mgx_status mgx_init(int device, mgx_context **out) {
 extern "C" mgx_status mgx_destroy(mgx_context *ctx) {
   if (!ctx) return MGX_OK;
   if (ctx->comm) (void)mgx_comm_destroy(ctx);
+  for (int i = 0; i < 2; ++i) {
+    if (ctx->louv_col[i]) (void)hipFree(ctx->louv_col[i]);
+    if (ctx->louv_w[i]) (void)hipFree(ctx->louv_w[i]);
+  }
   if (ctx->workspace) (void)ctx->free_async(ctx->workspace);
   ctx->cache_trim();
   for (auto &e : ctx->cache_live) (void)hipFree(e.first);  // leaked by callers

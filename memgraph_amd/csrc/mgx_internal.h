@@ -84,6 +84,18 @@ struct mgx_context {
   // Grow-only device workspace for rocPRIM temp storage etc.
   void *workspace = nullptr;
   size_t workspace_bytes = 0;
+  // Louvain coarsen col/w ping-pong arenas. The ROCm 7.0.x runtime
+  // misplaces >4 GiB hipMallocs once the VA space is fragmented
+  // (profiles/r02_summary.md), and the coarsen's total-sized col/w are
+  // the only such allocations issued mid-run — so they come from these
+  // grow-only arenas, sized and allocated at mgx_louvain entry before
+  // any per-level churn. [2] because level k's graph must stay live
+  // while level k+1 is being written.
+  void *louv_col[2] = {nullptr, nullptr};
+  void *louv_w[2] = {nullptr, nullptr};
+  size_t louv_col_bytes[2] = {0, 0};
+  size_t louv_w_bytes[2] = {0, 0};
+  int louv_flip = 0;
   mgx_comm_state *comm = nullptr;
 
   mgx_status reserve(size_t bytes, void **out);
